@@ -1,0 +1,94 @@
+"""Public API surface of the kubeshare-amd stack: pod labels, annotations,
+environment variables, on-disk paths and operational defaults.
+
+The label vocabulary is the reference's public workload API and is kept
+verbatim (reference: pkg/scheduler/constants.go:3-28, README.md:32-105).
+Everything NVIDIA-specific is replaced by the ROCm-native equivalent
+(env injection: reference pkg/scheduler/pod.go:435-476).
+"""
+
+# ---------------------------------------------------------------- labels (L5)
+DOMAIN = "sharedgpu/"
+
+POD_GROUP_NAME = DOMAIN + "group_name"
+POD_GROUP_HEADCOUNT = DOMAIN + "group_headcount"
+POD_GROUP_THRESHOLD = DOMAIN + "group_threshold"
+# The aggregator vocabulary for gang size; the reference accepts it in
+# aggregator/pod.go:22 while the scheduler derives it from
+# headcount*threshold (pod_group.go:86-117). We accept both (SURVEY.md
+# Appendix A) and document min_available as the canonical exported form.
+POD_MIN_AVAILABLE = DOMAIN + "min_available"
+
+POD_PRIORITY = DOMAIN + "priority"
+POD_GPU_LIMIT = DOMAIN + "gpu_limit"
+POD_GPU_REQUEST = DOMAIN + "gpu_request"
+POD_GPU_MEMORY = DOMAIN + "gpu_mem"
+POD_GPU_MODEL = DOMAIN + "gpu_model"
+
+# annotations written by Reserve (reference pod.go:402-476)
+POD_GPU_UUID = DOMAIN + "gpu_uuid"
+POD_CELL_ID = DOMAIN + "cell_id"
+POD_MANAGER_PORT = DOMAIN + "gpu_manager_port"
+
+SCHEDULER_NAME = "kubeshare-scheduler"
+
+# ------------------------------------------------------------- env vars (L1)
+# Injected into shared-GPU containers (ROCm-native replacements for the
+# reference's NVIDIA_VISIBLE_DEVICES / LD_PRELOAD block, pod.go:445-457).
+ENV_ROCR_VISIBLE_DEVICES = "ROCR_VISIBLE_DEVICES"
+ENV_HIP_VISIBLE_DEVICES = "HIP_VISIBLE_DEVICES"
+ENV_LD_PRELOAD = "LD_PRELOAD"
+ENV_POD_MANAGER_IP = "POD_MANAGER_IP"
+ENV_POD_MANAGER_PORT = "POD_MANAGER_PORT"
+ENV_POD_NAME = "POD_NAME"
+ENV_SCHEDULER_IP = "SCHEDULER_IP"
+ENV_SCHEDULER_PORT = "SCHEDULER_PORT"
+# memory cap in bytes enforced by libhiphook (default request * full memory)
+ENV_GPU_MEM = "KUBESHARE_GPU_MEM"
+# abort instead of silently running un-gated when the hook failed to attach
+ENV_REQUIRE_HOOK = "KUBESHARE_REQUIRE_HOOK"
+
+# ------------------------------------------------------------ hostPath (L2)
+KUBESHARE_ROOT = "/kubeshare"
+LIBRARY_PATH = KUBESHARE_ROOT + "/library"
+HOOK_SO_NAME = "libhiphook.so"
+HOOK_SO_PATH = LIBRARY_PATH + "/" + HOOK_SO_NAME
+SCHEDULER_IP_FILE = LIBRARY_PATH + "/schedulerIP.txt"
+LOG_PATH = KUBESHARE_ROOT + "/log"
+SCHEDULER_CONFIG_ROOT = KUBESHARE_ROOT + "/scheduler"
+GPU_CONFIG_DIR = SCHEDULER_CONFIG_ROOT + "/config/"
+POD_MANAGER_PORT_DIR = SCHEDULER_CONFIG_ROOT + "/podmanagerport/"
+CLUSTER_TOPOLOGY_FILE = SCHEDULER_CONFIG_ROOT + "/kubeshare-config.yaml"
+
+# ------------------------------------------------------------- defaults (L1)
+# Token-scheduler knobs; values are the reference's operational defaults
+# (docker/kubeshare-gemini-scheduler/launcher.py:77-80).
+BASE_QUOTA_MS = 300.0
+MIN_QUOTA_MS = 20.0
+WINDOW_MS = 10000.0
+
+# gpu-schd listens on BASE_SCHED_PORT + gpu_index
+# (launcher-multigpus.sh:21,41); pod managers draw from a 512-wide pool
+# (pkg/scheduler/node.go:13-15, scheduler.go:348-360).
+BASE_SCHED_PORT = 49901
+POD_MANAGER_PORT_START = 50050
+POD_MANAGER_PORT_POOL = 512
+
+# gang semantics (reference scheduler.go:44-47)
+PERMIT_WAITING_TIME_SEC = 2  # * group headcount
+POD_GROUP_GC_INTERVAL_SEC = 30
+POD_GROUP_EXPIRATION_SEC = 600
+
+# --------------------------------------------------------------- MI355X (HW)
+MI355X_MODEL = "AMD Instinct MI355X"
+MI355X_GFX = "gfx950"
+MI355X_HBM_BYTES = 288 * 1024**3  # 288 GiB HBM3E
+MI355X_GPUS_PER_NODE = 8
+MI355X_XGMI_LINKS_PER_GPU = 7  # point-to-point, ~153 GB/s each
+MI355X_XGMI_LINK_GBPS = 153.0
+
+# ------------------------------------------------------------- telemetry (L3)
+METRIC_GPU_CAPACITY = "gpu_capacity"       # reference pkg/collector/collector.go:30-35
+METRIC_GPU_REQUIREMENT = "gpu_requirement"  # reference pkg/aggregator/aggregator.go:22-39
+COLLECTOR_PORT = 9004
+AGGREGATOR_PORT = 9005

@@ -1,0 +1,267 @@
+// gpu-schd core: per-GPU sliding-window token scheduler (policy only,
+// no I/O — unit-testable; the poll loop lives in gpu_schd.cpp).
+//
+// Role model: the reference's gem-schd (one instance per physical GPU,
+// CLI flags -q 300 -m 20 -w 10000, SURVEY.md §2.2 and
+// docker/kubeshare-gemini-scheduler/launcher.py:27-31,77-80). Internals
+// are re-derived: deficit-based work-conserving time slicing.
+//
+// Model
+// -----
+// At most ONE token is outstanding per GPU. A token is a wall-time lease
+// of quota_ms during which the holder may submit GPU work freely; the
+// holder drains its streams at expiry and reports measured use (RET).
+// Co-located pods therefore run in alternating near-exclusive windows —
+// on MI355X there is no enforced SR-IOV partitioning inside one GPU, so
+// time-multiplexing the 256 CUs at ~300 ms granularity is both the
+// simplest and the highest-throughput isolation mechanism (each slice
+// runs with full HBM bandwidth and all CUs; no cache thrash from
+// co-running kernels).
+//
+// Fairness policy (per grant decision):
+//   usage(pod)  = sum of charged ms inside the sliding window w
+//   over(pod)   = usage >= limit*w            -> ineligible (hard cap)
+//   deficit     = request*w - usage           -> most under-served wins
+//   tie         = lowest usage/limit          -> work-conserving burst
+//   quota       = clamp(min(base_q, limit*w - usage), min_q, base_q)
+//
+// Guarantee: a pod with gpu_request=r that is continuously hungry gets
+// >= r of the window (its deficit dominates as soon as it drops below
+// r*w); a pod alone on the GPU gets up to limit*w (work conservation).
+#pragma once
+
+#include <algorithm>
+#include <deque>
+#include <map>
+#include <string>
+#include <vector>
+
+#include "../common/protocol.hpp"
+
+namespace ks {
+
+struct PodAccount {
+  double request = 0.0;
+  double limit = 1.0;
+  long long memory = 0;       // byte cap (0 = default)
+  bool in_config = false;     // listed in the per-UUID config file?
+  // (t_charge_end_ms, used_ms) entries inside the window
+  std::deque<std::pair<double, double>> charges;
+  double usage_cache = 0.0;
+  // lifetime stats
+  double total_used_ms = 0.0;
+  long long grants = 0;
+  long long mem_bytes = 0;    // last reported footprint
+
+  void prune(double now, double window_ms) {
+    while (!charges.empty() && charges.front().first < now - window_ms) {
+      usage_cache -= charges.front().second;
+      charges.pop_front();
+    }
+    if (charges.empty()) usage_cache = 0.0;
+  }
+  void charge(double now, double used_ms) {
+    charges.emplace_back(now, used_ms);
+    usage_cache += used_ms;
+    total_used_ms += used_ms;
+  }
+};
+
+struct Waiter {
+  std::string pod;
+  long long cookie;   // opaque per-request id assigned by the I/O layer
+  double hint_ms;
+  double enq_ms;
+};
+
+struct Grant {
+  std::string pod;
+  long long cookie = -1;
+  double quota_ms = 0.0;
+};
+
+class TokenScheduler {
+ public:
+  TokenScheduler(double base_quota_ms, double min_quota_ms, double window_ms)
+      : base_q_(base_quota_ms), min_q_(min_quota_ms), window_(window_ms) {}
+
+  void set_config(const std::vector<PodQuota>& quotas) {
+    for (auto& kv : pods_) kv.second.in_config = false;
+    for (const auto& q : quotas) {
+      PodAccount& a = pods_[q.pod];
+      a.request = q.request;
+      a.limit = q.limit;
+      a.memory = q.memory;
+      a.in_config = true;
+    }
+  }
+
+  PodAccount& account(const std::string& pod) {
+    auto it = pods_.find(pod);
+    if (it != pods_.end()) return it->second;
+    // Unknown pod (config file may lag pod start — the reference's
+    // Prometheus round-trip had the same gap): run it opportunistically
+    // with no guarantee and full burst until the config lands.
+    PodAccount a;
+    a.request = 0.0;
+    a.limit = 1.0;
+    pods_[pod] = a;
+    return pods_[pod];
+  }
+
+  // A pod asked for a token.
+  void request(const std::string& pod, long long cookie, double hint_ms,
+               double now) {
+    waiters_.push_back(Waiter{pod, cookie, hint_ms, now});
+  }
+
+  // The holder returned its token. `used_ms` is the client-measured GPU
+  // time; it is clamped server-side to [0, 3x quota + slack] so a lying
+  // or buggy client cannot starve others by under- or over-reporting
+  // less than it actually blocked the GPU (wall enforcement below
+  // catches over-holding).
+  void release(const std::string& pod, double used_ms, double now) {
+    if (!holder_.empty() && pod == holder_) {
+      double wall = now - grant_ms_;
+      double charged = std::max(0.0, std::min(used_ms, wall + 50.0));
+      // a GPU-bound holder's wall time is the true exclusive occupancy
+      charged = std::max(charged, std::min(wall, grant_quota_));
+      account(pod).charge(now, charged);
+      holder_.clear();
+    } else {
+      // late RET after a revocation: already charged, ignore amount
+    }
+  }
+
+  // Holder liveness: revoke a token whose holder exceeded its lease by
+  // revoke_factor (it died or hung; its connection may still be open).
+  bool check_revoke(double now, double* revoked_at = nullptr) {
+    if (holder_.empty()) return false;
+    double deadline = grant_ms_ + grant_quota_ * 3.0 + 2000.0;
+    if (now < deadline) return false;
+    account(holder_).charge(now, now - grant_ms_);
+    if (revoked_at) *revoked_at = now;
+    holder_.clear();
+    return true;
+  }
+
+  void drop_pod_waiters(long long cookie_lo, long long cookie_hi) {
+    waiters_.erase(std::remove_if(waiters_.begin(), waiters_.end(),
+                                  [&](const Waiter& w) {
+                                    return w.cookie >= cookie_lo &&
+                                           w.cookie <= cookie_hi;
+                                  }),
+                   waiters_.end());
+  }
+
+  // If a holder vanished without RET (connection closed), charge wall.
+  void force_release(const std::string& pod, double now) {
+    if (holder_ == pod) {
+      account(pod).charge(now, now - grant_ms_);
+      holder_.clear();
+    }
+  }
+
+  // Pick the next holder. Returns true and fills `out` when a token was
+  // granted; when false and `next_try_ms` > 0, every waiter is over its
+  // hard cap and the caller should re-run at now+next_try_ms.
+  bool schedule(double now, Grant* out, double* next_try_ms) {
+    *next_try_ms = 0.0;
+    if (!holder_.empty() || waiters_.empty()) return false;
+
+    double best_key = -1e300;
+    int best_i = -1;
+    double soonest = 1e300;
+    for (int i = 0; i < (int)waiters_.size(); i++) {
+      PodAccount& a = account(waiters_[i].pod);
+      a.prune(now, window_);
+      double cap = a.limit * window_;
+      if (a.usage_cache >= cap - 1e-9) {
+        // over hard cap: eligible again when the oldest charge ages out
+        if (!a.charges.empty())
+          soonest = std::min(soonest,
+                             a.charges.front().first + window_ - now);
+        continue;
+      }
+      double deficit = a.request * window_ - a.usage_cache;
+      // primary: deficit (guarantee); secondary: burst fairness
+      double key = deficit > 0.0
+                       ? 1e6 + deficit
+                       : -(a.usage_cache / std::max(a.limit, 1e-9));
+      // FIFO tiebreak: earlier enqueue wins on exact ties
+      key -= (waiters_[i].enq_ms - now) * 1e-12;
+      if (key > best_key) {
+        best_key = key;
+        best_i = i;
+      }
+    }
+    if (best_i < 0) {
+      *next_try_ms = std::max(1.0, std::min(soonest, window_));
+      return false;
+    }
+
+    Waiter w = waiters_[best_i];
+    waiters_.erase(waiters_.begin() + best_i);
+    PodAccount& a = account(w.pod);
+    double room = a.limit * window_ - a.usage_cache;
+    double quota = std::min(base_q_, room);
+    quota = std::max(quota, min_q_);
+    holder_ = w.pod;
+    grant_ms_ = now;
+    grant_quota_ = quota;
+    a.grants++;
+    out->pod = w.pod;
+    out->cookie = w.cookie;
+    out->quota_ms = quota;
+    return true;
+  }
+
+  const std::string& holder() const { return holder_; }
+  size_t n_waiters() const { return waiters_.size(); }
+
+  // One-line JSON stats: per-pod window usage, share of busy time,
+  // quota error vs request. Used by bench.py and the GPU tests as the
+  // *server-side* quota-enforcement measurement.
+  std::string stats_json(double now) {
+    double total_busy = 0.0;
+    for (auto& kv : pods_) {
+      kv.second.prune(now, window_);
+      total_busy += kv.second.usage_cache;
+    }
+    std::string s = "{\"window_ms\":" + fmt(window_) +
+                    ",\"busy_ms\":" + fmt(total_busy) + ",\"pods\":{";
+    bool first = true;
+    for (auto& kv : pods_) {
+      const PodAccount& a = kv.second;
+      if (!first) s += ",";
+      first = false;
+      double share = total_busy > 0 ? a.usage_cache / total_busy : 0.0;
+      double frac = a.usage_cache / window_;
+      s += "\"" + kv.first + "\":{\"request\":" + fmt(a.request) +
+           ",\"limit\":" + fmt(a.limit) + ",\"usage_ms\":" +
+           fmt(a.usage_cache) + ",\"window_frac\":" + fmt(frac) +
+           ",\"busy_share\":" + fmt(share) + ",\"grants\":" +
+           std::to_string(a.grants) + ",\"total_used_ms\":" +
+           fmt(a.total_used_ms) + ",\"mem_bytes\":" +
+           std::to_string(a.mem_bytes) + "}";
+    }
+    s += "}}";
+    return s;
+  }
+
+ private:
+  static std::string fmt(double v) {
+    char b[32];
+    snprintf(b, sizeof(b), "%.4f", v);
+    return b;
+  }
+
+  double base_q_, min_q_, window_;
+  std::map<std::string, PodAccount> pods_;
+  std::vector<Waiter> waiters_;
+  std::string holder_;
+  double grant_ms_ = 0.0;
+  double grant_quota_ = 0.0;
+};
+
+}  // namespace ks

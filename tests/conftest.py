@@ -1,0 +1,31 @@
+import os
+import shutil
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+NATIVE = os.path.join(REPO, "native")
+sys.path.insert(0, REPO)
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires a real MI355X (run via gpurun)")
+
+
+def _built(name: str) -> str:
+    return os.path.join(NATIVE, name)
+
+
+@pytest.fixture(scope="session")
+def native_bins():
+    """Build (if needed) and return paths of the native daemons."""
+    targets = ["gpu-schd", "pod-mgr", "hook_selftest", "libhiphook.so"]
+    if not all(os.path.exists(_built(t)) for t in targets):
+        if shutil.which("make") is None:
+            pytest.skip("make unavailable")
+        subprocess.run(["make", "-C", NATIVE], check=True,
+                       capture_output=True)
+    return {t: _built(t) for t in targets}
