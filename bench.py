@@ -1,0 +1,208 @@
+#!/usr/bin/env python3
+"""bench.py — the BASELINE.json headline benchmark.
+
+Metric: aggregate images/sec + quota-error % for N co-located pods at
+gpu_request=0.5 per MI355X. Per GPU (= per rank) this stands up the full
+production isolation chain — gpu-schd, one pod-mgr per pod, and TWO pod
+processes with LD_PRELOAD=libhiphook.so — each training the flagship
+model (ResNet50, bf16 autocast, channels-last, synthetic data,
+random-init weights) for exactly --steps full training steps.
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+`value` is the WHOLE-JOB aggregate images/s over all pods on all N GPUs;
+`ms_per_step` is the max-over-ranks time for the K steps of the
+co-located pair. Quota enforcement error (server-side, from gpu-schd's
+sliding-window accounting) is reported in config.quota_error_pct.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import subprocess
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+
+def ensure_native():
+    from kubeshare_amd.isolation.local import NATIVE_DIR
+    needed = ["gpu-schd", "pod-mgr", "libhiphook.so"]
+    if not all(os.path.exists(os.path.join(NATIVE_DIR, n)) for n in needed):
+        subprocess.run(["make", "-C", NATIVE_DIR], check=True,
+                       capture_output=True)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--model", default="resnet50")
+    ap.add_argument("--batch", type=int, default=96,
+                    help="per-pod batch size")
+    ap.add_argument("--image-size", type=int, default=224)
+    ap.add_argument("--pods-per-gpu", type=int, default=2)
+    ap.add_argument("--request", type=float, default=0.5)
+    ap.add_argument("--limit", type=float, default=1.0)
+    ap.add_argument("--dtype", default="bf16")
+    ap.add_argument("--device", default=None,
+                    help="override (cpu for plumbing tests)")
+    ap.add_argument("--use-ops", default="auto")
+    args = ap.parse_args()
+
+    import torch
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    distributed = world > 1
+
+    on_gpu = torch.cuda.is_available() and args.device != "cpu"
+    device = args.device or (f"cuda:{local_rank}" if on_gpu else "cpu")
+
+    dist = None
+    if distributed:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        backend = "nccl" if on_gpu else "gloo"
+        if on_gpu:
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend=backend)
+
+    ensure_native()
+    from kubeshare_amd.isolation.local import LocalGPUShare
+
+    share = LocalGPUShare(gpu_index=local_rank,
+                          full_memory=(torch.cuda.get_device_properties(
+                              local_rank).total_memory if on_gpu
+                              else 288 * 1024**3))
+    share.start()
+
+    workers = []
+    try:
+        for i in range(args.pods_per_gpu):
+            pod_name = f"bench/pod{local_rank}-{i}"
+            h = share.add_pod(pod_name, request=args.request,
+                              limit=args.limit)
+            env = h.env(gpu_index=local_rank)
+            if not on_gpu:
+                env.pop("KUBESHARE_REQUIRE_HOOK", None)
+            # the pod sees exactly one GPU (ROCR_VISIBLE_DEVICES) -> cuda:0
+            wdev = "cuda:0" if on_gpu else "cpu"
+            cmd = [sys.executable, "-m", "kubeshare_amd.bench_worker",
+                   "--model", args.model, "--batch", str(args.batch),
+                   "--image-size", str(args.image_size),
+                   "--steps", str(args.steps), "--warmup", str(args.warmup),
+                   "--device", wdev, "--dtype",
+                   "bf16" if args.dtype == "bf16" else "fp32",
+                   "--use-ops", args.use_ops]
+            env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+            p = subprocess.Popen(cmd, env=env, cwd=REPO,
+                                 stdin=subprocess.PIPE,
+                                 stdout=subprocess.PIPE, text=True,
+                                 bufsize=1)
+            workers.append((pod_name, p))
+
+        # wait for warmup on every pod of this rank
+        for name, p in workers:
+            line = p.stdout.readline().strip()
+            if line != "READY":
+                raise RuntimeError(f"{name}: expected READY, got {line!r} "
+                                   f"(exit={p.poll()})")
+
+        # ---- timed region: barrier + device sync on both sides ----
+        if dist:
+            dist.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _, p in workers:
+            p.stdin.write("GO\n")
+            p.stdin.flush()
+        results = []
+        for name, p in workers:
+            line = p.stdout.readline().strip()
+            if not line.startswith("DONE"):
+                raise RuntimeError(f"{name}: expected DONE, got {line!r}")
+            _, elapsed_s, images, loss = line.split()
+            results.append((float(elapsed_s), int(images), float(loss)))
+        if on_gpu:
+            torch.cuda.synchronize()
+        t1 = time.perf_counter()
+        if dist:
+            dist.barrier()
+        elapsed = t1 - t0
+
+        quota_err = share.quota_error_pct()
+        stats = share.stats()
+    finally:
+        for _, p in workers:
+            if p.poll() is None:
+                p.kill()
+        share.stop()
+
+    rank_images = sum(r[1] for r in results)
+
+    if dist:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        n = torch.tensor([float(rank_images)], dtype=torch.float64)
+        if on_gpu:
+            t, n = t.cuda(), n.cuda()
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        dist.all_reduce(n, op=dist.ReduceOp.SUM)
+        max_elapsed = float(t.item())
+        total_images = float(n.item())
+    else:
+        max_elapsed = elapsed
+        total_images = float(rank_images)
+
+    if rank == 0:
+        pods_total = args.pods_per_gpu * world
+        mean_qerr = (sum(quota_err.values()) / len(quota_err)
+                     if quota_err else None)
+        out = {
+            "metric": "aggregate images/sec, co-located shared-GPU pods",
+            "value": round(total_images / max_elapsed, 2),
+            "unit": "images/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(max_elapsed / args.steps * 1000.0, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * pods_total,
+                "image_size": args.image_size,
+                "parallelism": f"{pods_total} pods @ request="
+                               f"{args.request}, {args.pods_per_gpu}/GPU",
+                "pods_per_gpu": args.pods_per_gpu,
+                "gpu_request": args.request,
+                "gpu_limit": args.limit,
+                "quota_error_pct": round(mean_qerr, 3)
+                                    if mean_qerr is not None else None,
+                "quota_error_per_pod": {k: round(v, 3)
+                                        for k, v in quota_err.items()},
+                "sched_window_ms": stats.get("window_ms"),
+                "per_pod_loss": [r[2] for r in results],
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+    if dist:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,189 @@
+"""GPU tests (MI355X): the isolation chain under real HIP traffic.
+
+Run via: gpurun -- 'python -m pytest tests -m gpu -x -q'
+"""
+import ctypes
+import os
+import subprocess
+import sys
+import time
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+torch = pytest.importorskip("torch")
+if not torch.cuda.is_available():
+    pytest.skip("no GPU", allow_module_level=True)
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.fixture(scope="module")
+def share(native_bins):
+    from kubeshare_amd.isolation.local import LocalGPUShare
+    s = LocalGPUShare(gpu_index=0, base_quota_ms=100, window_ms=4000)
+    s.start()
+    yield s
+    s.stop()
+
+
+def _spawn_burner(handle, duration_ms, extra_env=None):
+    env = handle.env(gpu_index=0)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    if extra_env:
+        env.update(extra_env)
+    return subprocess.Popen(
+        [sys.executable, "-m", "kubeshare_amd.isolation.burn_worker",
+         "--duration-ms", str(duration_ms)],
+        env=env, cwd=REPO, stdout=subprocess.PIPE, text=True)
+
+
+def test_smoke_entry():
+    sys.path.insert(0, REPO)
+    import __graft_entry__ as g
+    g.smoke()
+
+
+def test_burn_wall_calibration():
+    """ks_ops.burn(ms) occupies the GPU for ~ms (wall_clock64 at 100MHz)."""
+    from kubeshare_amd import ops
+    ops.burn(5.0)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    ops.burn(80.0)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) * 1000
+    assert 60 < dt < 200, f"burn(80ms) took {dt:.1f}ms"
+
+
+def test_sgd_momentum_matches_torch():
+    """Numerics: fused HIP SGD vs plain PyTorch fp32 reference."""
+    from kubeshare_amd import ops
+    torch.manual_seed(0)
+    shapes = [(1000,), (64, 64), (3, 3, 17), (2048, 1000)]
+    ps = [torch.randn(s, device="cuda") for s in shapes]
+    gs = [torch.randn(s, device="cuda") for s in shapes]
+    ref_ps = [p.clone() for p in ps]
+
+    # reference: torch.optim.SGD
+    for p, g in zip(ref_ps, gs):
+        p.grad = g.clone()
+    opt = torch.optim.SGD(ref_ps, lr=0.1, momentum=0.9, weight_decay=1e-4)
+    for _ in range(3):
+        opt.step()
+
+    fused = ops.FusedSGD(
+        [p.requires_grad_() for p in ps], lr=0.1, momentum=0.9,
+        weight_decay=1e-4)
+    for p, g in zip(ps, gs):
+        p.grad = g.clone()
+    for _ in range(3):
+        fused.step()
+    torch.cuda.synchronize()
+    for p, r in zip(ps, ref_ps):
+        torch.testing.assert_close(p, r, rtol=1e-5, atol=1e-6)
+
+
+def test_two_pods_5050_split(share):
+    """Config #2 of BASELINE.json: 2 pods @0.5 on one MI355X, ~50/50
+    kernel-time split measured server-side."""
+    a = share.add_pod("gpu/a", request=0.5, limit=1.0)
+    b = share.add_pod("gpu/b", request=0.5, limit=1.0)
+    try:
+        pa = _spawn_burner(a, 8000)
+        pb = _spawn_burner(b, 8000)
+        out_a, _ = pa.communicate(timeout=120)
+        out_b, _ = pb.communicate(timeout=120)
+        assert pa.returncode == 0 and pb.returncode == 0, (out_a, out_b)
+        ua = float(out_a.split()[4])
+        ub = float(out_b.split()[4])
+        assert ua > 0 and ub > 0
+        share_a = ua / (ua + ub)
+        assert abs(share_a - 0.5) < 0.12, f"hook-side split {share_a}"
+        st = share.stats()
+        sa = st["pods"]["gpu/a"]["busy_share"]
+        assert abs(sa - 0.5) < 0.12, f"schd-side split {sa}"
+        leases_a = int(out_a.split()[3])
+        assert leases_a > 3, "token gating never engaged"
+    finally:
+        share.remove_pod("gpu/a")
+        share.remove_pod("gpu/b")
+
+
+def test_asymmetric_split_75_25(share):
+    a = share.add_pod("gpu/big", request=0.75, limit=0.75)
+    b = share.add_pod("gpu/small", request=0.25, limit=0.25)
+    try:
+        pa = _spawn_burner(a, 8000)
+        pb = _spawn_burner(b, 8000)
+        out_a, _ = pa.communicate(timeout=120)
+        out_b, _ = pb.communicate(timeout=120)
+        ua = float(out_a.split()[4])
+        ub = float(out_b.split()[4])
+        share_a = ua / (ua + ub)
+        assert abs(share_a - 0.75) < 0.12, f"split {share_a}"
+    finally:
+        share.remove_pod("gpu/big")
+        share.remove_pod("gpu/small")
+
+
+def test_memory_cap_enforced(share):
+    """hipMalloc beyond KUBESHARE_GPU_MEM must fail -> torch OOM; and
+    mem_get_info must report the clamped capacity."""
+    h = share.add_pod("gpu/mem", request=0.5, limit=1.0,
+                      memory=2 * 1024**3)
+    try:
+        code = (
+            "import torch, json;"
+            "free,total = torch.cuda.mem_get_info();"
+            "assert total <= 2*1024**3 + (1<<20), f'total {total}';"
+            "ok=False\n"
+            "try:\n"
+            "    x = torch.empty(4*1024**3, dtype=torch.uint8, device='cuda')\n"
+            "except torch.cuda.OutOfMemoryError:\n"
+            "    ok=True\n"
+            "assert ok, 'allocation over cap succeeded'\n"
+            "y = torch.empty(512*1024**2, dtype=torch.uint8, device='cuda')\n"
+            "print('MEMCAP-OK')\n"
+        )
+        env = h.env(gpu_index=0)
+        env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+        r = subprocess.run([sys.executable, "-c", code], env=env, cwd=REPO,
+                           capture_output=True, text=True, timeout=300)
+        assert "MEMCAP-OK" in r.stdout, (r.stdout, r.stderr)
+    finally:
+        share.remove_pod("gpu/mem")
+
+
+def test_hook_fail_loud_without_preload(share):
+    """KUBESHARE_REQUIRE_HOOK=1 without LD_PRELOAD: the worker must
+    refuse to run (no silent un-isolated execution)."""
+    h = share.add_pod("gpu/loud", request=0.5)
+    try:
+        env = h.env(gpu_index=0)
+        env.pop("LD_PRELOAD")
+        env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+        r = subprocess.run(
+            [sys.executable, "-m", "kubeshare_amd.bench_worker",
+             "--model", "resnet18", "--batch", "2", "--image-size", "64",
+             "--steps", "1", "--warmup", "0"],
+            env=env, cwd=REPO, capture_output=True, text=True, timeout=300)
+        assert r.returncode != 0
+        assert "libhiphook not attached" in r.stderr
+    finally:
+        share.remove_pod("gpu/loud")
+
+
+def test_work_conserving_solo_pod(share):
+    """One pod with request 0.5 / limit 1.0 and no competitor should get
+    nearly the whole GPU (burst-to-limit)."""
+    h = share.add_pod("gpu/solo", request=0.5, limit=1.0)
+    try:
+        p = _spawn_burner(h, 5000)
+        out, _ = p.communicate(timeout=120)
+        wall = float(out.split()[1])
+        used = float(out.split()[4])
+        assert used > 0.75 * wall * 1000, f"burst throttled: {out}"
+    finally:
+        share.remove_pod("gpu/solo")
