@@ -58,10 +58,33 @@ namespace {
 using ks::now_ms;
 
 // ------------------------------------------------------------ real symbols
+// RTLD_NEXT only searches the GLOBAL scope after this library; PyTorch
+// loads libamdhip64 as a dependency of an RTLD_LOCAL-dlopened module,
+// so RTLD_NEXT can miss it — fall back to the runtime's own handle
+// (already loaded: dlopen just bumps the refcount, and dlsym(handle)
+// reads libamdhip64's own table, not our preempting wrappers).
+inline void* hip_runtime_handle() {
+  static void* h = [] {
+    for (const char* n :
+         {"libamdhip64.so.7", "libamdhip64.so.6", "libamdhip64.so"}) {
+      if (void* x = dlopen(n, RTLD_NOW | RTLD_LOCAL | RTLD_NOLOAD)) return x;
+    }
+    for (const char* n :
+         {"libamdhip64.so.7", "libamdhip64.so.6", "libamdhip64.so"}) {
+      if (void* x = dlopen(n, RTLD_NOW | RTLD_LOCAL)) return x;
+    }
+    return (void*)nullptr;
+  }();
+  return h;
+}
+
 template <typename T>
 T real_sym(const char* name) {
   static_assert(sizeof(T) == sizeof(void*), "fn ptr");
   void* p = dlsym(RTLD_NEXT, name);
+  if (!p) {
+    if (void* h = hip_runtime_handle()) p = dlsym(h, name);
+  }
   return reinterpret_cast<T>(p);
 }
 
