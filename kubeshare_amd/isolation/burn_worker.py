@@ -15,6 +15,8 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--duration-ms", type=float, default=5000)
     ap.add_argument("--burn-ms", type=float, default=5.0)
+    ap.add_argument("--wait-go", action="store_true",
+                    help="print READY after warmup and wait for GO on stdin")
     args = ap.parse_args()
 
     import torch
@@ -24,6 +26,11 @@ def main():
     torch.cuda.init()
     ops.burn(1.0)
     torch.cuda.synchronize()
+
+    if args.wait_go:
+        print("READY", flush=True)
+        if not sys.stdin.readline().startswith("GO"):
+            return 1
 
     t0 = time.perf_counter()
     iters = 0

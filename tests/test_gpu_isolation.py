@@ -28,15 +28,28 @@ def share(native_bins):
     s.stop()
 
 
-def _spawn_burner(handle, duration_ms, extra_env=None):
+def _spawn_burner(handle, duration_ms, extra_env=None, wait_go=False):
     env = handle.env(gpu_index=0)
     env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
     if extra_env:
         env.update(extra_env)
-    return subprocess.Popen(
-        [sys.executable, "-m", "kubeshare_amd.isolation.burn_worker",
-         "--duration-ms", str(duration_ms)],
-        env=env, cwd=REPO, stdout=subprocess.PIPE, text=True)
+    cmd = [sys.executable, "-m", "kubeshare_amd.isolation.burn_worker",
+           "--duration-ms", str(duration_ms)]
+    if wait_go:
+        cmd.append("--wait-go")
+    return subprocess.Popen(cmd, env=env, cwd=REPO, stdin=subprocess.PIPE,
+                            stdout=subprocess.PIPE, text=True, bufsize=1)
+
+
+def _start_together(procs, timeout=180):
+    """Wait for READY from every burner, then send GO simultaneously
+    (removes torch-init skew from the measurement window)."""
+    for p in procs:
+        line = p.stdout.readline().strip()
+        assert line == "READY", f"expected READY, got {line!r}"
+    for p in procs:
+        p.stdin.write("GO\n")
+        p.stdin.flush()
 
 
 def test_smoke_entry():
@@ -91,8 +104,11 @@ def test_two_pods_5050_split(share):
     a = share.add_pod("gpu/a", request=0.5, limit=1.0)
     b = share.add_pod("gpu/b", request=0.5, limit=1.0)
     try:
-        pa = _spawn_burner(a, 8000)
-        pb = _spawn_burner(b, 8000)
+        pa = _spawn_burner(a, 8000, wait_go=True)
+        pb = _spawn_burner(b, 8000, wait_go=True)
+        _start_together([pa, pb])
+        time.sleep(6.0)
+        mid = share.stats()  # mid-run: window fully inside the overlap
         out_a, _ = pa.communicate(timeout=120)
         out_b, _ = pb.communicate(timeout=120)
         assert pa.returncode == 0 and pb.returncode == 0, (out_a, out_b)
@@ -101,8 +117,7 @@ def test_two_pods_5050_split(share):
         assert ua > 0 and ub > 0
         share_a = ua / (ua + ub)
         assert abs(share_a - 0.5) < 0.12, f"hook-side split {share_a}"
-        st = share.stats()
-        sa = st["pods"]["gpu/a"]["busy_share"]
+        sa = mid["pods"]["gpu/a"]["busy_share"]
         assert abs(sa - 0.5) < 0.12, f"schd-side split {sa}"
         leases_a = int(out_a.split()[3])
         assert leases_a > 3, "token gating never engaged"
@@ -115,8 +130,9 @@ def test_asymmetric_split_75_25(share):
     a = share.add_pod("gpu/big", request=0.75, limit=0.75)
     b = share.add_pod("gpu/small", request=0.25, limit=0.25)
     try:
-        pa = _spawn_burner(a, 8000)
-        pb = _spawn_burner(b, 8000)
+        pa = _spawn_burner(a, 8000, wait_go=True)
+        pb = _spawn_burner(b, 8000, wait_go=True)
+        _start_together([pa, pb])
         out_a, _ = pa.communicate(timeout=120)
         out_b, _ = pb.communicate(timeout=120)
         ua = float(out_a.split()[4])
