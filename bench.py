@@ -45,7 +45,7 @@ def main():
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--model", default="resnet50")
-    ap.add_argument("--batch", type=int, default=96,
+    ap.add_argument("--batch", type=int, default=256,
                     help="per-pod batch size")
     ap.add_argument("--image-size", type=int, default=224)
     ap.add_argument("--pods-per-gpu", type=int, default=2)

@@ -51,13 +51,14 @@ def main():
     ap.add_argument("--device", default="cuda:0")
     ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     ap.add_argument("--no-channels-last", action="store_true")
-    ap.add_argument("--lr", type=float, default=0.1)
+    ap.add_argument("--lr", type=float, default=0.02)
     ap.add_argument("--use-ops", default="auto", choices=["auto", "on", "off"],
                     help="use the kubeshare_amd HIP fused ops")
     args = ap.parse_args()
 
     import torch
 
+    torch.backends.cudnn.benchmark = True  # MIOpen find+cache per shape
     dev = torch.device(args.device)
     on_gpu = dev.type == "cuda"
     if on_gpu:

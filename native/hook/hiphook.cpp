@@ -221,7 +221,7 @@ inline void gate2(void* ra, hipStream_t stream) {
     last_ex = exempt_caller(ra);
   }
   if (last_ex) return;
-  if (s.gate.lease_valid()) return;
+  if (s.gate.touch_lease_valid()) return;
   if (stream_capturing(stream)) return;  // capture: no GPU time, no sync
   s.gate.ensure();
 }

@@ -8,6 +8,11 @@
 // never had.
 //
 //   hook_selftest <host|uds> <port> <pod> <duration_ms> [stats]
+//   hook_selftest <host|uds> <port> <pod> <duration_ms> gate <active_ms> <idle_ms>
+//
+// The second form drives the REAL TokenGate (the object libhiphook uses)
+// with a stub drain, alternating bursts of simulated dispatches with
+// idle phases — exercising lease renewal and the idle-release watchdog.
 //
 // Prints one line: "DONE <pod> <leases> <granted_ms>" and, with `stats`,
 // the scheduler's STATS JSON line.
@@ -15,9 +20,30 @@
 
 using namespace ks;
 
+static int run_gate_mode(const char* host, int port, const std::string& pod,
+                         double duration, double active_ms, double idle_ms) {
+  static TokenGate gate;
+  gate.init(host, port, pod, [](void*) { /* GPU already drained */ },
+            nullptr);
+  double t0 = now_ms();
+  while (now_ms() - t0 < duration) {
+    double burst_end = now_ms() + active_ms;
+    while (now_ms() < burst_end && now_ms() - t0 < duration) {
+      gate.ensure();  // a "kernel launch" every ~1 ms
+      usleep(1000);
+    }
+    if (idle_ms > 0) usleep((useconds_t)(idle_ms * 1000));
+  }
+  gate.relinquish();
+  printf("DONE %s %lld %.1f idle_releases=%lld\n", pod.c_str(),
+         gate.leases(), gate.used_ms_total(), gate.idle_releases_);
+  return 0;
+}
+
 int main(int argc, char** argv) {
   if (argc < 5) {
-    fprintf(stderr, "usage: hook_selftest <host> <port> <pod> <duration_ms> [stats]\n");
+    fprintf(stderr, "usage: hook_selftest <host> <port> <pod> <duration_ms> "
+                    "[stats | gate <active_ms> <idle_ms>]\n");
     return 2;
   }
   const char* host = argv[1];
@@ -25,6 +51,9 @@ int main(int argc, char** argv) {
   std::string pod = argv[3];
   double duration = atof(argv[4]);
   bool want_stats = argc >= 6 && strcmp(argv[5], "stats") == 0;
+  if (argc >= 8 && strcmp(argv[5], "gate") == 0)
+    return run_gate_mode(host, port, pod, duration, atof(argv[6]),
+                         atof(argv[7]));
 
   TokenClient client;
   client.configure(host, port, pod);

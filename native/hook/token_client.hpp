@@ -120,6 +120,13 @@ class TokenClient {
 // The gate: fast-path check shared by every hooked entry point.
 // drain() is injected so this stays HIP-free (hiphook.cpp passes
 // hipDeviceSynchronize; tests pass a stub that models GPU work).
+//
+// Idle release: a lease holder that stops dispatching (end of job,
+// CPU-bound phase, waiting on input) must not sit on the token until
+// expiry/revocation — that would idle the GPU for everyone. A watchdog
+// thread relinquishes the lease once no gated call has arrived for
+// idle_release_ms (drains first, so queued work is charged), making the
+// slicing truly work-conserving; the next dispatch re-acquires.
 class TokenGate {
  public:
   using DrainFn = void (*)(void*);
@@ -129,20 +136,27 @@ class TokenGate {
     client_.configure(host, port, pod);
     drain_ = drain;
     drain_arg_ = drain_arg;
+    if (const char* e = getenv("KUBESHARE_IDLE_RELEASE_MS"))
+      idle_release_ms_ = atof(e);
     enabled_.store(client_.configured(), std::memory_order_release);
   }
 
   bool enabled() const { return enabled_.load(std::memory_order_acquire); }
 
-  // True while the current lease covers `now` (the dispatch fast path).
-  inline bool lease_valid() const {
-    return now_ms() < deadline_.load(std::memory_order_acquire);
+  // The dispatch fast path: records activity (for the idle watchdog)
+  // and reports whether the current lease covers `now`.
+  inline bool touch_lease_valid() {
+    double now = now_ms();
+    last_activity_.store(now, std::memory_order_relaxed);
+    return now < deadline_.load(std::memory_order_acquire);
   }
 
-  // The hot path: ~2 ns when the lease is valid.
+  // The hot path: ~ns when the lease is valid.
   inline void ensure() {
     if (!enabled()) return;
-    if (now_ms() < deadline_.load(std::memory_order_acquire)) return;
+    double now = now_ms();
+    last_activity_.store(now, std::memory_order_relaxed);
+    if (now < deadline_.load(std::memory_order_acquire)) return;
     renew();
   }
 
@@ -165,10 +179,12 @@ class TokenGate {
   }
 
   void reset_after_fork() {
-    // mutex state may be inconsistent post-fork; re-init
+    // mutex state may be inconsistent post-fork; re-init. The watchdog
+    // thread does not survive fork — restarted on the next acquire.
     pthread_mutex_init(&mu_, nullptr);
     client_.reset_after_fork();
     holding_ = false;
+    watchdog_running_.store(false, std::memory_order_release);
     deadline_.store(-1.0, std::memory_order_release);
   }
 
@@ -176,6 +192,58 @@ class TokenGate {
   double used_ms_total() const { return client_.returned_ms(); }
 
  private:
+  static void* watchdog_entry(void* self) {
+    static_cast<TokenGate*>(self)->watchdog_loop();
+    return nullptr;
+  }
+
+  void watchdog_loop() {
+    const double tick = std::max(5.0, idle_release_ms_ / 4.0);
+    for (;;) {
+      usleep((useconds_t)(tick * 1000));
+      if (!enabled()) continue;
+      double dl = deadline_.load(std::memory_order_acquire);
+      if (dl < 0) continue;  // no lease
+      double idle = now_ms() - last_activity_.load(std::memory_order_relaxed);
+      if (idle < idle_release_ms_) continue;
+      // No dispatches for a while: give the token back (drains first,
+      // so still-running queued work is fully charged).
+      pthread_mutex_lock(&mu_);
+      if (holding_ &&
+          now_ms() - last_activity_.load(std::memory_order_relaxed) >=
+              idle_release_ms_) {
+        double t_pre = now_ms();
+        if (drain_) drain_(drain_arg_);
+        double t_post = now_ms();
+        double used;
+        if (t_post - t_pre < 2.0) {
+          // GPU was already idle: charge only up to last activity +
+          // the idle threshold, not the watchdog's reaction time
+          used = std::min(
+              t_post,
+              last_activity_.load(std::memory_order_relaxed) +
+                  idle_release_ms_) - grant_time_;
+        } else {
+          used = t_post - grant_time_;  // drain did real work
+        }
+        client_.return_token(std::max(0.0, used));
+        holding_ = false;
+        deadline_.store(-1.0, std::memory_order_release);
+        idle_releases_++;
+      }
+      pthread_mutex_unlock(&mu_);
+    }
+  }
+
+  void start_watchdog_locked() {
+    if (watchdog_running_.load(std::memory_order_acquire)) return;
+    pthread_t t;
+    if (pthread_create(&t, nullptr, watchdog_entry, this) == 0) {
+      pthread_detach(t);
+      watchdog_running_.store(true, std::memory_order_release);
+    }
+  }
+
   void renew() {
     pthread_mutex_lock(&mu_);
     double now = now_ms();
@@ -191,6 +259,7 @@ class TokenGate {
       client_.return_token(used);
       holding_ = false;
     }
+    start_watchdog_locked();
     double quota = client_.acquire(0.0);
     if (quota <= 0.0) {
       // Scheduler unreachable: fail OPEN for liveness (the node daemon
@@ -203,6 +272,7 @@ class TokenGate {
       return;
     }
     grant_time_ = now_ms();
+    last_activity_.store(grant_time_, std::memory_order_relaxed);
     deadline_.store(grant_time_ + quota, std::memory_order_release);
     holding_ = true;
     pthread_mutex_unlock(&mu_);
@@ -213,12 +283,16 @@ class TokenGate {
   void* drain_arg_ = nullptr;
   pthread_mutex_t mu_ = PTHREAD_MUTEX_INITIALIZER;
   std::atomic<bool> enabled_{false};
+  std::atomic<bool> watchdog_running_{false};
   std::atomic<double> deadline_{-1.0};
+  std::atomic<double> last_activity_{0.0};
+  double idle_release_ms_ = 25.0;
   bool holding_ = false;
   double grant_time_ = 0.0;
 
  public:
   long long failures_ = 0;
+  long long idle_releases_ = 0;
 };
 
 }  // namespace ks

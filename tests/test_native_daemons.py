@@ -252,3 +252,34 @@ def test_dead_holder_is_revoked(native_bins, tmp_path):
         assert res["ns/b"][1] > 500, "token not released after holder death"
     finally:
         schd.stop()
+
+
+def test_idle_release_work_conserving(native_bins, tmp_path):
+    """A bursty pod (100ms active / 400ms idle) must hand the GPU to a
+    continuous competitor during its idle phases via the hook watchdog,
+    not stall it until the liveness revoke."""
+    cfg = _write_config(tmp_path, [("ns/bursty", 1.0, 0.5, 0),
+                                   ("ns/greedy", 1.0, 0.5, 0)])
+    schd = Schd(native_bins, cfg, q=150, m=10, w=4000)
+    try:
+        pa = subprocess.Popen(
+            [native_bins["hook_selftest"], "127.0.0.1", str(schd.port),
+             "ns/bursty", "4000", "gate", "100", "400"],
+            stdout=subprocess.PIPE, text=True)
+        pb = subprocess.Popen(
+            [native_bins["hook_selftest"], "127.0.0.1", str(schd.port),
+             "ns/greedy", "4000", "gate", "4000", "0"],
+            stdout=subprocess.PIPE, text=True)
+        out_a, _ = pa.communicate(timeout=60)
+        out_b, _ = pb.communicate(timeout=60)
+        assert pa.returncode == 0 and pb.returncode == 0
+        used_a = float(out_a.split()[3])
+        used_b = float(out_b.split()[3])
+        idle_rel = int(out_a.split()[4].split("=")[1])
+        assert idle_rel >= 2, f"watchdog never released: {out_a}"
+        # greedy pod picks up bursty's idle time (work conservation)
+        assert used_b > 2200, (out_a, out_b)
+        # bursty still makes progress during its active phases
+        assert used_a > 400, (out_a, out_b)
+    finally:
+        schd.stop()
