@@ -39,6 +39,23 @@ def ensure_native():
                        capture_output=True)
 
 
+def _quota_error(st0: dict, st1: dict) -> dict:
+    """Per-pod |achieved GPU-time share - entitled share| in % over the
+    timed region, from gpu-schd's cumulative per-pod counters (server-
+    side: a pod cannot fake it)."""
+    used = {}
+    for pod, v in st1.get("pods", {}).items():
+        before = st0.get("pods", {}).get(pod, {}).get("total_used_ms", 0.0)
+        used[pod] = max(0.0, v["total_used_ms"] - before)
+    total = sum(used.values())
+    if total <= 0:
+        return {}
+    reqs = {p: st1["pods"][p]["request"] for p in used}
+    total_req = sum(reqs.values()) or 1.0
+    return {p: abs(used[p] / total - reqs[p] / total_req) * 100.0
+            for p in used}
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -122,6 +139,7 @@ def main():
             dist.barrier()
         if on_gpu:
             torch.cuda.synchronize()
+        st0 = share.stats()  # cumulative per-pod GPU ms before the region
         t0 = time.perf_counter()
         for _, p in workers:
             p.stdin.write("GO\n")
@@ -140,8 +158,11 @@ def main():
             dist.barrier()
         elapsed = t1 - t0
 
-        quota_err = share.quota_error_pct()
+        # let the hook watchdogs RET the final leases so the cumulative
+        # counters include the tail of the timed region
+        time.sleep(0.4)
         stats = share.stats()
+        quota_err = _quota_error(st0, stats)
     finally:
         for _, p in workers:
             if p.poll() is None:

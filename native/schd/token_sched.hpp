@@ -175,8 +175,12 @@ class TokenScheduler {
     for (int i = 0; i < (int)waiters_.size(); i++) {
       PodAccount& a = account(waiters_[i].pod);
       a.prune(now, window_);
+      // limit >= 1.0 means "may use the whole GPU": no hard cap. (A
+      // busy solo pod's window usage approaches window_ by definition;
+      // clamping it to ever-smaller leases would thrash on drains.)
+      bool uncapped = a.limit >= 0.999;
       double cap = a.limit * window_;
-      if (a.usage_cache >= cap - 1e-9) {
+      if (!uncapped && a.usage_cache >= cap - 1e-9) {
         // over hard cap: eligible again when the oldest charge ages out
         if (!a.charges.empty())
           soonest = std::min(soonest,
@@ -203,7 +207,8 @@ class TokenScheduler {
     Waiter w = waiters_[best_i];
     waiters_.erase(waiters_.begin() + best_i);
     PodAccount& a = account(w.pod);
-    double room = a.limit * window_ - a.usage_cache;
+    double room = a.limit >= 0.999 ? base_q_
+                                   : a.limit * window_ - a.usage_cache;
     double quota = std::min(base_q_, room);
     quota = std::max(quota, min_q_);
     holder_ = w.pod;
