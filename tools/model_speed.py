@@ -16,7 +16,7 @@ sys.path.insert(0, REPO)
 
 
 def run_cfg(model_name, batch, channels_last, benchmark, dtype, steps,
-            warmup):
+            warmup, graph=False):
     import torch
     from kubeshare_amd.models import build_model
 
@@ -32,15 +32,47 @@ def run_cfg(model_name, batch, channels_last, benchmark, dtype, steps,
     y = torch.randint(0, 1000, (batch,), device=dev)
     amp = torch.bfloat16 if dtype == "bf16" else None
 
-    def step():
-        opt.zero_grad(set_to_none=True)
+    def fwd_bwd():
         if amp:
             with torch.autocast("cuda", dtype=amp):
                 loss = torch.nn.functional.cross_entropy(model(x), y)
         else:
             loss = torch.nn.functional.cross_entropy(model(x), y)
         loss.backward()
+        return loss
+
+    def step():
+        opt.zero_grad(set_to_none=True)
+        fwd_bwd()
         opt.step()
+
+    g = None
+    if graph:
+        # hipGraph-captured whole step: zero_grad must keep buffers
+        # (static addresses) and warmup must run on a side stream
+        for _ in range(max(3, warmup)):
+            step()
+        opt.zero_grad(set_to_none=False)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                for p in model.parameters():
+                    p.grad.zero_()
+                fwd_bwd()
+                opt.step()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        with torch.cuda.graph(g):
+            for p in model.parameters():
+                p.grad.zero_()
+            fwd_bwd()
+            opt.step()
+
+        def step():  # noqa: F811
+            g.replay()
 
     for _ in range(warmup):
         step()
@@ -50,11 +82,11 @@ def run_cfg(model_name, batch, channels_last, benchmark, dtype, steps,
         step()
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
-    del model, opt, x, y
+    del model, opt, x, y, g
     torch.cuda.empty_cache()
     return {
         "model": model_name, "batch": batch, "channels_last": channels_last,
-        "benchmark": benchmark, "dtype": dtype,
+        "benchmark": benchmark, "dtype": dtype, "graph": graph,
         "ms_per_step": round(dt / steps * 1000, 2),
         "images_per_s": round(batch * steps / dt, 1),
     }
@@ -63,26 +95,37 @@ def run_cfg(model_name, batch, channels_last, benchmark, dtype, steps,
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--quick", action="store_true")
+    ap.add_argument("--graphs", action="store_true",
+                    help="compare eager vs hipGraph-captured step")
+    ap.add_argument("--one", action="store_true",
+                    help="single canonical config (for rocprof)")
+    ap.add_argument("--batch", type=int, default=256)
     ap.add_argument("--model", default="resnet50")
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=6)
     args = ap.parse_args()
 
-    if args.quick:
+    if args.one:
+        cfgs = [(args.batch, True, True, "bf16", False)]
+    elif args.graphs:
+        cfgs = [(args.batch, True, True, "bf16", False),
+                (args.batch, True, True, "bf16", True),
+                (512, True, True, "bf16", True)]
+    elif args.quick:
         grid = [(256, True, True), (256, False, True), (256, True, False),
                 (64, True, True), (512, True, True)]
-        cfgs = [(b, cl, bm, "bf16") for b, cl, bm in grid]
+        cfgs = [(b, cl, bm, "bf16", False) for b, cl, bm in grid]
     else:
-        cfgs = [(b, cl, bm, d) for b, cl, bm, d in itertools.product(
+        cfgs = [(b, cl, bm, d, False) for b, cl, bm, d in itertools.product(
             [64, 128, 256, 512], [True, False], [True, False], ["bf16"])]
 
-    for batch, cl, bm, dtype in cfgs:
+    for batch, cl, bm, dtype, graph in cfgs:
         try:
             r = run_cfg(args.model, batch, cl, bm, dtype, args.steps,
-                        args.warmup)
+                        args.warmup, graph=graph)
         except RuntimeError as e:
             r = {"batch": batch, "channels_last": cl, "benchmark": bm,
-                 "error": str(e)[:200]}
+                 "graph": graph, "error": str(e)[:300]}
         print(json.dumps(r), flush=True)
 
 
