@@ -56,6 +56,9 @@ def main():
                     help="use the kubeshare_amd HIP fused ops")
     args = ap.parse_args()
 
+    from kubeshare_amd.utils.tuning import apply_miopen_tuning
+    apply_miopen_tuning()  # before the first conv
+
     import torch
 
     torch.backends.cudnn.benchmark = True  # MIOpen find+cache per shape

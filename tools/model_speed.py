@@ -17,6 +17,8 @@ sys.path.insert(0, REPO)
 
 def run_cfg(model_name, batch, channels_last, benchmark, dtype, steps,
             warmup, graph=False):
+    from kubeshare_amd.utils.tuning import apply_miopen_tuning
+    apply_miopen_tuning()
     import torch
     from kubeshare_amd.models import build_model
 

@@ -18,6 +18,8 @@ def main():
     ap.add_argument("--steps", type=int, default=5)
     args = ap.parse_args()
 
+    from kubeshare_amd.utils.tuning import apply_miopen_tuning
+    apply_miopen_tuning()
     import torch
     from torch.profiler import ProfilerActivity, profile
 
