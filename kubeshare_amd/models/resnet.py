@@ -15,6 +15,7 @@ import torch.nn as nn
 
 class Bottleneck(nn.Module):
     expansion = 4
+    fused_ops = False  # set by kubeshare_amd.ops.fuse_model
 
     def __init__(self, in_ch: int, width: int, stride: int = 1):
         super().__init__()
@@ -36,6 +37,11 @@ class Bottleneck(nn.Module):
 
     def forward(self, x):
         identity = x if self.down is None else self.down(x)
+        if self.fused_ops:
+            from .. import ops
+            out = ops.bn_relu(self.conv1(x), self.bn1)
+            out = ops.bn_relu(self.conv2(out), self.bn2)
+            return ops.bn_relu(self.conv3(out), self.bn3, res=identity)
         out = self.relu(self.bn1(self.conv1(x)))
         out = self.relu(self.bn2(self.conv2(out)))
         out = self.bn3(self.conv3(out))
@@ -44,6 +50,7 @@ class Bottleneck(nn.Module):
 
 class BasicBlock(nn.Module):
     expansion = 1
+    fused_ops = False  # set by kubeshare_amd.ops.fuse_model
 
     def __init__(self, in_ch: int, width: int, stride: int = 1):
         super().__init__()
@@ -62,12 +69,18 @@ class BasicBlock(nn.Module):
 
     def forward(self, x):
         identity = x if self.down is None else self.down(x)
+        if self.fused_ops:
+            from .. import ops
+            out = ops.bn_relu(self.conv1(x), self.bn1)
+            return ops.bn_relu(self.conv2(out), self.bn2, res=identity)
         out = self.relu(self.bn1(self.conv1(x)))
         out = self.bn2(self.conv2(out))
         return self.relu(out + identity)
 
 
 class ResNet(nn.Module):
+    fused_ops = False  # stem BN+ReLU fusion (set by ops.fuse_model)
+
     def __init__(self, block, layers, num_classes: int = 1000):
         super().__init__()
         self.in_ch = 64
@@ -96,7 +109,11 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        if self.fused_ops:
+            from .. import ops
+            x = self.maxpool(ops.bn_relu(self.conv1(x), self.bn1))
+        else:
+            x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = torch.flatten(self.avgpool(x), 1)
         return self.fc(x)

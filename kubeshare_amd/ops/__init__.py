@@ -11,7 +11,8 @@ import os
 
 _HERE = os.path.dirname(os.path.abspath(__file__))
 _BUILD_DIR = os.path.join(_HERE, "_build")
-_SRC = os.path.join(_HERE, "hip", "ks_ops.hip")
+_SRCS = [os.path.join(_HERE, "hip", "ks_ops.hip"),
+         os.path.join(_HERE, "hip", "bn_relu.hip")]
 
 _ext = None
 
@@ -26,7 +27,7 @@ def build_extension(verbose: bool = False):
     os.makedirs(_BUILD_DIR, exist_ok=True)
     return cpp_extension.load(
         name="ks_ops",
-        sources=[_SRC],
+        sources=_SRCS,
         build_directory=_BUILD_DIR,
         extra_cflags=["-O3"],
         verbose=verbose,
@@ -108,9 +109,93 @@ class FusedSGD:
                           self.weight_decay)
 
 
+# ---------------------------------------------------- fused BN+ReLU(+add)
+def _bn_relu_autograd():
+    """Lazily build the autograd.Function (torch import deferred)."""
+    global _BNReLUFn
+    if _BNReLUFn is not None:
+        return _BNReLUFn
+    import torch
+
+    class BNReLUFn(torch.autograd.Function):
+        """y = relu(batch_norm(x) [+ res]) on NHWC bf16 via the gfx950
+        kernels in hip/bn_relu.hip. Replaces MIOpen's 5-pass forward /
+        8-pass backward (incl. separate ReLU and residual-add kernels)
+        with 3 / 7 passes — see profiles/README.md for the motivation."""
+
+        @staticmethod
+        def forward(ctx, x, res, weight, bias, running_mean, running_var,
+                    training, momentum, eps):
+            ext = _load()
+            w32 = weight.float()
+            b32 = bias.float()
+            if training:
+                y, mean, invstd = ext.bn_relu_fwd_train(
+                    x, w32, b32, running_mean, running_var, momentum, eps,
+                    res)
+                ctx.save_for_backward(x, y, w32, mean, invstd)
+                ctx.with_res = res is not None
+                ctx.wdtype = weight.dtype
+                return y
+            y = ext.bn_relu_fwd_eval(x, w32, b32, running_mean.float(),
+                                     running_var.float(), eps, res)
+            ctx.save_for_backward(x, y, w32, running_mean.float(),
+                                  (running_var.float() + eps).rsqrt())
+            ctx.with_res = res is not None
+            ctx.wdtype = weight.dtype
+            return y
+
+        @staticmethod
+        def backward(ctx, dy):
+            ext = _load()
+            x, y, w32, mean, invstd = ctx.saved_tensors
+            out = ext.bn_relu_bwd(x, y, dy, w32, mean, invstd, ctx.with_res)
+            dx, dscale, dbias = out[0], out[1], out[2]
+            dres = out[3] if ctx.with_res else None
+            return (dx, dres, dscale.to(ctx.wdtype), dbias.to(ctx.wdtype),
+                    None, None, None, None, None)
+
+    _BNReLUFn = BNReLUFn
+    return BNReLUFn
+
+
+_BNReLUFn = None
+
+
+def bn_relu(x, bn, res=None):
+    """Fused BN+ReLU (+residual add) using an nn.BatchNorm2d's
+    parameters/buffers; falls back to eager for shapes the kernel does
+    not cover (non-bf16, C%8!=0, not channels-last)."""
+    import torch
+
+    supported = (x.dtype == torch.bfloat16 and x.size(1) % 8 == 0
+                 and x.size(1) <= 2048 and x.is_cuda
+                 and x.is_contiguous(memory_format=torch.channels_last))
+    if supported and res is not None and res.dtype != torch.bfloat16:
+        # downsample-path BN runs in fp32 under autocast; the residual
+        # join is bf16 in bf16 training
+        res = res.to(torch.bfloat16)
+    if not supported:
+        y = torch.nn.functional.batch_norm(
+            x, bn.running_mean, bn.running_var, bn.weight, bn.bias,
+            bn.training, bn.momentum, bn.eps)
+        if res is not None:
+            y = y + res
+        return torch.relu(y)
+    if res is not None and not res.is_contiguous(
+            memory_format=torch.channels_last):
+        res = res.contiguous(memory_format=torch.channels_last)
+    fn = _bn_relu_autograd()
+    return fn.apply(x, res, bn.weight, bn.bias, bn.running_mean,
+                    bn.running_var, bn.training, bn.momentum, bn.eps)
+
+
 def fuse_model(model):
-    """Swap fusable modules for HIP-fused versions. v1: verifies the
-    extension is importable (fail-loud on GPU); module-level fusions
-    (BN+ReLU) land on top of this hook."""
+    """Enable the fused BN+ReLU(+add) path on kubeshare_amd models
+    (ResNet blocks check their `fused_ops` flag); verifies the extension
+    is importable first (fail-loud on GPU boxes)."""
     _load()
+    for m in model.modules():
+        if hasattr(m, "fused_ops"):
+            m.fused_ops = True
     return model

@@ -1,0 +1,466 @@
+// Fused NHWC bf16 BatchNorm + ReLU (+ residual add) for gfx950.
+//
+// Motivation (profiles/resnet50_bs256_kernel_breakdown_untuned.txt):
+// MIOpen's spatial BN kernels + separate ReLU/add elementwise passes
+// are ~48% of the non-conv GPU time of a ResNet50 training step. The
+// stock path makes 5 full passes over the activation in forward
+// (mean/var read, normalize read+write, clamp read+write) and 8 in
+// backward; these kernels make 3 and 7 (and fold the residual add and
+// its gradient for free).
+//
+// Design (cdna_hip_programming.md G13, Appx B "Reduction"):
+//  - channels-last bf16, C % 8 == 0: one lane owns 8 CONSECUTIVE
+//    channels (one 16-byte load), so per-channel reductions never cross
+//    lanes — each lane accumulates f32x8 partials over its rows, a
+//    block-level LDS tree reduces the lanes that share a channel group,
+//    and one global f32 atomicAdd per channel per block merges blocks.
+//  - wave64-friendly: blockDim 256, lane -> (channel_group, row_offset)
+//    fixed for the whole kernel, so scale/bias live in registers.
+//  - all stats/parameter math in f32; tensors read/written as bf16x8.
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <vector>
+
+typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
+typedef __attribute__((ext_vector_type(8))) float float8;
+
+__device__ __forceinline__ float bf16_to_f32(unsigned short u) {
+  union {
+    unsigned int i;
+    float f;
+  } v;
+  v.i = ((unsigned int)u) << 16;
+  return v.f;
+}
+
+__device__ __forceinline__ unsigned short f32_to_bf16(float f) {
+  union {
+    float f;
+    unsigned int i;
+  } v;
+  v.f = f;
+  // round-to-nearest-even (matches PyTorch's float->bf16 cast)
+  unsigned int lsb = (v.i >> 16) & 1;
+  v.i += 0x7fffu + lsb;
+  return (unsigned short)(v.i >> 16);
+}
+
+#define KS_BN_BLOCK 256
+
+// ------------------------------------------------------------- fwd stats
+// sum[c] += x, sumsq[c] += x^2 over all M rows. Lane l owns channel
+// group cg = l % CG (8 channels), row offset l / CG.
+__global__ void bn_stats_kernel(const ushort8* __restrict__ x,
+                                float* __restrict__ sum,
+                                float* __restrict__ sumsq,
+                                long long M, int CG) {
+  __shared__ float s_sum[KS_BN_BLOCK * 8];
+  __shared__ float s_sq[KS_BN_BLOCK * 8];
+  const int tid = threadIdx.x;
+  const int cg = tid % CG;
+  const int roff = tid / CG;
+  const int rows_per_blk = KS_BN_BLOCK / CG;
+
+  float8 acc = {0, 0, 0, 0, 0, 0, 0, 0};
+  float8 acc2 = {0, 0, 0, 0, 0, 0, 0, 0};
+  long long row = (long long)blockIdx.x * rows_per_blk + roff;
+  const long long stride = (long long)gridDim.x * rows_per_blk;
+  for (; row < M; row += stride) {
+    ushort8 v = x[row * CG + cg];
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float f = bf16_to_f32(v[j]);
+      acc[j] += f;
+      acc2[j] += f * f;
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; j++) {
+    s_sum[tid * 8 + j] = acc[j];
+    s_sq[tid * 8 + j] = acc2[j];
+  }
+  __syncthreads();
+  // lanes 0..CG-1 reduce over the rows_per_blk lanes sharing their cg
+  if (tid < CG) {
+    float8 t = {0, 0, 0, 0, 0, 0, 0, 0};
+    float8 t2 = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int r = 0; r < rows_per_blk; r++) {
+      const int src = r * CG + tid;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        t[j] += s_sum[src * 8 + j];
+        t2[j] += s_sq[src * 8 + j];
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      atomicAdd(&sum[tid * 8 + j], t[j]);
+      atomicAdd(&sumsq[tid * 8 + j], t2[j]);
+    }
+  }
+}
+
+// ------------------------------------------------------- fwd finalize
+// mean/invstd from the reduced sums; running-stat update (PyTorch
+// semantics: running = (1-m)*running + m*batch, var unbiased); and the
+// folded per-channel scale' = w*invstd, bias' = b - mean*scale'.
+__global__ void bn_finalize_kernel(const float* __restrict__ sum,
+                                   const float* __restrict__ sumsq,
+                                   const float* __restrict__ weight,
+                                   const float* __restrict__ bias,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var,
+                                   float* __restrict__ save_mean,
+                                   float* __restrict__ save_invstd,
+                                   float* __restrict__ scale_out,
+                                   float* __restrict__ bias_out,
+                                   long long M, int C, float momentum,
+                                   float eps) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float mean = sum[c] / (float)M;
+  float var = sumsq[c] / (float)M - mean * mean;
+  var = var < 0.f ? 0.f : var;
+  float invstd = rsqrtf(var + eps);
+  save_mean[c] = mean;
+  save_invstd[c] = invstd;
+  if (running_mean) {
+    float unbiased = M > 1 ? var * (float)M / (float)(M - 1) : var;
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  }
+  float s = weight[c] * invstd;
+  scale_out[c] = s;
+  bias_out[c] = bias[c] - mean * s;
+}
+
+// --------------------------------------------------------- fwd apply
+// y = relu(x*scale' + bias' [+ res]); scale/bias in registers (lane's
+// channel group is fixed).
+template <bool WITH_RES>
+__global__ void bn_apply_relu_kernel(const ushort8* __restrict__ x,
+                                     const ushort8* __restrict__ res,
+                                     ushort8* __restrict__ y,
+                                     const float* __restrict__ scale,
+                                     const float* __restrict__ biasf,
+                                     long long M, int CG) {
+  const int tid = threadIdx.x;
+  const int cg = tid % CG;
+  const int roff = tid / CG;
+  const int rows_per_blk = KS_BN_BLOCK / CG;
+  float8 s, b;
+#pragma unroll
+  for (int j = 0; j < 8; j++) {
+    s[j] = scale[cg * 8 + j];
+    b[j] = biasf[cg * 8 + j];
+  }
+  long long row = (long long)blockIdx.x * rows_per_blk + roff;
+  const long long stride = (long long)gridDim.x * rows_per_blk;
+  for (; row < M; row += stride) {
+    const long long k = row * CG + cg;
+    ushort8 v = x[k];
+    ushort8 r;
+    if (WITH_RES) r = res[k];
+    ushort8 o;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float f = fmaf(bf16_to_f32(v[j]), s[j], b[j]);
+      if (WITH_RES) f += bf16_to_f32(r[j]);
+      o[j] = f32_to_bf16(f > 0.f ? f : 0.f);
+    }
+    y[k] = o;
+  }
+}
+
+// --------------------------------------------------------- bwd stats
+// dym = dy * (y > 0); dbias[c] = sum(dym); dscale[c] = sum(dym * xhat).
+// Optionally writes dym out (it IS the residual branch's gradient for
+// the BN+Add+ReLU fusion, and the apply pass re-reads it).
+template <bool WRITE_DYM>
+__global__ void bn_bwd_stats_kernel(const ushort8* __restrict__ x,
+                                    const ushort8* __restrict__ y,
+                                    const ushort8* __restrict__ dy,
+                                    ushort8* __restrict__ dym_out,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ invstd,
+                                    float* __restrict__ dbias,
+                                    float* __restrict__ dscale,
+                                    long long M, int CG) {
+  __shared__ float s_db[KS_BN_BLOCK * 8];
+  __shared__ float s_ds[KS_BN_BLOCK * 8];
+  const int tid = threadIdx.x;
+  const int cg = tid % CG;
+  const int roff = tid / CG;
+  const int rows_per_blk = KS_BN_BLOCK / CG;
+  float8 mu, is;
+#pragma unroll
+  for (int j = 0; j < 8; j++) {
+    mu[j] = mean[cg * 8 + j];
+    is[j] = invstd[cg * 8 + j];
+  }
+  float8 adb = {0, 0, 0, 0, 0, 0, 0, 0};
+  float8 ads = {0, 0, 0, 0, 0, 0, 0, 0};
+  long long row = (long long)blockIdx.x * rows_per_blk + roff;
+  const long long stride = (long long)gridDim.x * rows_per_blk;
+  for (; row < M; row += stride) {
+    const long long k = row * CG + cg;
+    ushort8 xv = x[k], yv = y[k], gv = dy[k];
+    ushort8 dm;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float g = bf16_to_f32(yv[j]) > 0.f ? bf16_to_f32(gv[j]) : 0.f;
+      float xhat = (bf16_to_f32(xv[j]) - mu[j]) * is[j];
+      adb[j] += g;
+      ads[j] += g * xhat;
+      if (WRITE_DYM) dm[j] = f32_to_bf16(g);
+    }
+    if (WRITE_DYM) dym_out[k] = dm;
+  }
+#pragma unroll
+  for (int j = 0; j < 8; j++) {
+    s_db[tid * 8 + j] = adb[j];
+    s_ds[tid * 8 + j] = ads[j];
+  }
+  __syncthreads();
+  if (tid < CG) {
+    float8 t = {0, 0, 0, 0, 0, 0, 0, 0};
+    float8 t2 = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int r = 0; r < rows_per_blk; r++) {
+      const int src = r * CG + tid;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        t[j] += s_db[src * 8 + j];
+        t2[j] += s_ds[src * 8 + j];
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      atomicAdd(&dbias[tid * 8 + j], t[j]);
+      atomicAdd(&dscale[tid * 8 + j], t2[j]);
+    }
+  }
+}
+
+// --------------------------------------------------------- bwd apply
+// dx = w*invstd * (dym - dbias/M - xhat * dscale/M)
+__global__ void bn_bwd_apply_kernel(const ushort8* __restrict__ x,
+                                    const ushort8* __restrict__ y,
+                                    const ushort8* __restrict__ dy,
+                                    ushort8* __restrict__ dx,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ invstd,
+                                    const float* __restrict__ weight,
+                                    const float* __restrict__ dbias,
+                                    const float* __restrict__ dscale,
+                                    long long M, int CG) {
+  const int tid = threadIdx.x;
+  const int cg = tid % CG;
+  const int roff = tid / CG;
+  const int rows_per_blk = KS_BN_BLOCK / CG;
+  const float invM = 1.f / (float)M;
+  float8 mu, is, w, db, ds;
+#pragma unroll
+  for (int j = 0; j < 8; j++) {
+    const int c = cg * 8 + j;
+    mu[j] = mean[c];
+    is[j] = invstd[c];
+    w[j] = weight[c] * is[j];
+    db[j] = dbias[c] * invM;
+    ds[j] = dscale[c] * invM;
+  }
+  long long row = (long long)blockIdx.x * rows_per_blk + roff;
+  const long long stride = (long long)gridDim.x * rows_per_blk;
+  for (; row < M; row += stride) {
+    const long long k = row * CG + cg;
+    ushort8 xv = x[k], yv = y[k], gv = dy[k];
+    ushort8 o;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float g = bf16_to_f32(yv[j]) > 0.f ? bf16_to_f32(gv[j]) : 0.f;
+      float xhat = (bf16_to_f32(xv[j]) - mu[j]) * is[j];
+      o[j] = f32_to_bf16(w[j] * (g - db[j] - xhat * ds[j]));
+    }
+    dx[k] = o;
+  }
+}
+
+// ---------------------------------------------------------- eval apply
+__global__ void bn_fold_eval_kernel(const float* __restrict__ weight,
+                                    const float* __restrict__ bias,
+                                    const float* __restrict__ rmean,
+                                    const float* __restrict__ rvar,
+                                    float* __restrict__ scale_out,
+                                    float* __restrict__ bias_out, int C,
+                                    float eps) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = weight[c] * rsqrtf(rvar[c] + eps);
+  scale_out[c] = s;
+  bias_out[c] = bias[c] - rmean[c] * s;
+}
+
+// ================================================================ host
+namespace {
+
+struct Geom {
+  long long M;
+  int C, CG, blocks;
+};
+
+Geom geom_of(const torch::Tensor& x) {
+  TORCH_CHECK(x.dim() == 4, "bn_relu: 4D NCHW tensor expected");
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "bn_relu: bf16 only");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "bn_relu: channels-last only");
+  Geom g;
+  g.C = (int)x.size(1);
+  TORCH_CHECK(g.C % 8 == 0 && g.C <= KS_BN_BLOCK * 8,
+              "bn_relu: C must be a multiple of 8 and <= 2048");
+  g.M = x.numel() / g.C;
+  g.CG = g.C / 8;
+  long long work = g.M * g.CG;
+  long long blocks = (work + KS_BN_BLOCK - 1) / KS_BN_BLOCK;
+  // G11: cap + grid-stride (256 CUs want >>256 workgroups)
+  g.blocks = (int)std::min<long long>(blocks, 2048);
+  return g;
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> bn_relu_fwd_train(
+    torch::Tensor x, torch::Tensor weight, torch::Tensor bias,
+    torch::Tensor running_mean, torch::Tensor running_var, double momentum,
+    double eps, c10::optional<torch::Tensor> res) {
+  Geom g = geom_of(x);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  auto f32 = x.options().dtype(at::kFloat);
+  auto sum = at::zeros({g.C * 2}, f32);
+  float* sum_p = sum.data_ptr<float>();
+  float* sumsq_p = sum_p + g.C;
+  auto y = at::empty_like(x);
+  auto save_mean = at::empty({g.C}, f32);
+  auto save_invstd = at::empty({g.C}, f32);
+  auto scale = at::empty({g.C}, f32);
+  auto biasf = at::empty({g.C}, f32);
+
+  hipLaunchKernelGGL(bn_stats_kernel, dim3(g.blocks), dim3(KS_BN_BLOCK), 0,
+                     stream.stream(),
+                     reinterpret_cast<const ushort8*>(x.data_ptr()), sum_p,
+                     sumsq_p, g.M, g.CG);
+  int fb = (g.C + 255) / 256;
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(fb), dim3(256), 0,
+                     stream.stream(), sum_p, sumsq_p,
+                     weight.data_ptr<float>(), bias.data_ptr<float>(),
+                     running_mean.defined()
+                         ? running_mean.data_ptr<float>() : nullptr,
+                     running_var.defined()
+                         ? running_var.data_ptr<float>() : nullptr,
+                     save_mean.data_ptr<float>(),
+                     save_invstd.data_ptr<float>(), scale.data_ptr<float>(),
+                     biasf.data_ptr<float>(), g.M, g.C, (float)momentum,
+                     (float)eps);
+  if (res.has_value()) {
+    hipLaunchKernelGGL(bn_apply_relu_kernel<true>, dim3(g.blocks),
+                       dim3(KS_BN_BLOCK), 0, stream.stream(),
+                       reinterpret_cast<const ushort8*>(x.data_ptr()),
+                       reinterpret_cast<const ushort8*>(res->data_ptr()),
+                       reinterpret_cast<ushort8*>(y.data_ptr()),
+                       scale.data_ptr<float>(), biasf.data_ptr<float>(), g.M,
+                       g.CG);
+  } else {
+    hipLaunchKernelGGL(bn_apply_relu_kernel<false>, dim3(g.blocks),
+                       dim3(KS_BN_BLOCK), 0, stream.stream(),
+                       reinterpret_cast<const ushort8*>(x.data_ptr()),
+                       nullptr, reinterpret_cast<ushort8*>(y.data_ptr()),
+                       scale.data_ptr<float>(), biasf.data_ptr<float>(), g.M,
+                       g.CG);
+  }
+  return {y, save_mean, save_invstd};
+}
+
+torch::Tensor bn_relu_fwd_eval(torch::Tensor x, torch::Tensor weight,
+                               torch::Tensor bias, torch::Tensor rmean,
+                               torch::Tensor rvar, double eps,
+                               c10::optional<torch::Tensor> res) {
+  Geom g = geom_of(x);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  auto f32 = x.options().dtype(at::kFloat);
+  auto scale = at::empty({g.C}, f32);
+  auto biasf = at::empty({g.C}, f32);
+  auto y = at::empty_like(x);
+  int fb = (g.C + 255) / 256;
+  hipLaunchKernelGGL(bn_fold_eval_kernel, dim3(fb), dim3(256), 0,
+                     stream.stream(), weight.data_ptr<float>(),
+                     bias.data_ptr<float>(), rmean.data_ptr<float>(),
+                     rvar.data_ptr<float>(), scale.data_ptr<float>(),
+                     biasf.data_ptr<float>(), g.C, (float)eps);
+  if (res.has_value()) {
+    hipLaunchKernelGGL(bn_apply_relu_kernel<true>, dim3(g.blocks),
+                       dim3(KS_BN_BLOCK), 0, stream.stream(),
+                       reinterpret_cast<const ushort8*>(x.data_ptr()),
+                       reinterpret_cast<const ushort8*>(res->data_ptr()),
+                       reinterpret_cast<ushort8*>(y.data_ptr()),
+                       scale.data_ptr<float>(), biasf.data_ptr<float>(), g.M,
+                       g.CG);
+  } else {
+    hipLaunchKernelGGL(bn_apply_relu_kernel<false>, dim3(g.blocks),
+                       dim3(KS_BN_BLOCK), 0, stream.stream(),
+                       reinterpret_cast<const ushort8*>(x.data_ptr()),
+                       nullptr, reinterpret_cast<ushort8*>(y.data_ptr()),
+                       scale.data_ptr<float>(), biasf.data_ptr<float>(), g.M,
+                       g.CG);
+  }
+  return y;
+}
+
+std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor x, torch::Tensor y,
+                                       torch::Tensor dy, torch::Tensor weight,
+                                       torch::Tensor mean,
+                                       torch::Tensor invstd, bool need_dres) {
+  Geom g = geom_of(x);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  auto f32 = x.options().dtype(at::kFloat);
+  auto red = at::zeros({g.C * 2}, f32);
+  float* dbias_p = red.data_ptr<float>();
+  float* dscale_p = dbias_p + g.C;
+  auto dx = at::empty_like(x);
+  torch::Tensor dres;
+  if (!dy.is_contiguous(at::MemoryFormat::ChannelsLast))
+    dy = dy.contiguous(at::MemoryFormat::ChannelsLast);
+
+  if (need_dres) {
+    dres = at::empty_like(x);
+    hipLaunchKernelGGL(bn_bwd_stats_kernel<true>, dim3(g.blocks),
+                       dim3(KS_BN_BLOCK), 0, stream.stream(),
+                       reinterpret_cast<const ushort8*>(x.data_ptr()),
+                       reinterpret_cast<const ushort8*>(y.data_ptr()),
+                       reinterpret_cast<const ushort8*>(dy.data_ptr()),
+                       reinterpret_cast<ushort8*>(dres.data_ptr()),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       dbias_p, dscale_p, g.M, g.CG);
+  } else {
+    hipLaunchKernelGGL(bn_bwd_stats_kernel<false>, dim3(g.blocks),
+                       dim3(KS_BN_BLOCK), 0, stream.stream(),
+                       reinterpret_cast<const ushort8*>(x.data_ptr()),
+                       reinterpret_cast<const ushort8*>(y.data_ptr()),
+                       reinterpret_cast<const ushort8*>(dy.data_ptr()),
+                       nullptr, mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), dbias_p, dscale_p, g.M,
+                       g.CG);
+  }
+  hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(g.blocks), dim3(KS_BN_BLOCK),
+                     0, stream.stream(),
+                     reinterpret_cast<const ushort8*>(x.data_ptr()),
+                     reinterpret_cast<const ushort8*>(y.data_ptr()),
+                     reinterpret_cast<const ushort8*>(dy.data_ptr()),
+                     reinterpret_cast<ushort8*>(dx.data_ptr()),
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     weight.data_ptr<float>(), dbias_p, dscale_p, g.M, g.CG);
+  auto dbias = red.narrow(0, 0, g.C);
+  auto dscale = red.narrow(0, g.C, g.C);
+  if (need_dres)
+    return {dx, dscale, dbias, dres};
+  return {dx, dscale, dbias};
+}

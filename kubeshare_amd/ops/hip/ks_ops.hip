@@ -113,9 +113,26 @@ void sgd_momentum_(std::vector<torch::Tensor> params,
   }
 }
 
+// fused NHWC bf16 BatchNorm+ReLU (bn_relu.hip)
+std::vector<torch::Tensor> bn_relu_fwd_train(
+    torch::Tensor x, torch::Tensor weight, torch::Tensor bias,
+    torch::Tensor running_mean, torch::Tensor running_var, double momentum,
+    double eps, c10::optional<torch::Tensor> res);
+torch::Tensor bn_relu_fwd_eval(torch::Tensor x, torch::Tensor weight,
+                               torch::Tensor bias, torch::Tensor rmean,
+                               torch::Tensor rvar, double eps,
+                               c10::optional<torch::Tensor> res);
+std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor x, torch::Tensor y,
+                                       torch::Tensor dy, torch::Tensor weight,
+                                       torch::Tensor mean,
+                                       torch::Tensor invstd, bool need_dres);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("burn", &burn, "occupy the GPU for ~ms milliseconds",
         py::arg("ms"), py::arg("blocks") = 1024, py::arg("threads") = 256);
   m.def("sgd_momentum_", &sgd_momentum_,
         "fused SGD+momentum update (in-place)");
+  m.def("bn_relu_fwd_train", &bn_relu_fwd_train);
+  m.def("bn_relu_fwd_eval", &bn_relu_fwd_eval);
+  m.def("bn_relu_bwd", &bn_relu_bwd);
 }

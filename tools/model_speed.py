@@ -16,7 +16,7 @@ sys.path.insert(0, REPO)
 
 
 def run_cfg(model_name, batch, channels_last, benchmark, dtype, steps,
-            warmup, graph=False):
+            warmup, graph=False, fused=False):
     from kubeshare_amd.utils.tuning import apply_miopen_tuning
     apply_miopen_tuning()
     import torch
@@ -24,7 +24,11 @@ def run_cfg(model_name, batch, channels_last, benchmark, dtype, steps,
 
     torch.backends.cudnn.benchmark = benchmark
     dev = torch.device("cuda:0")
-    model = build_model(model_name).to(dev)
+    model = build_model(model_name)
+    if fused:
+        from kubeshare_amd import ops
+        model = ops.fuse_model(model)
+    model = model.to(dev)
     if channels_last:
         model = model.to(memory_format=torch.channels_last)
     opt = torch.optim.SGD(model.parameters(), lr=0.02, momentum=0.9)
@@ -89,6 +93,7 @@ def run_cfg(model_name, batch, channels_last, benchmark, dtype, steps,
     return {
         "model": model_name, "batch": batch, "channels_last": channels_last,
         "benchmark": benchmark, "dtype": dtype, "graph": graph,
+        "fused": fused,
         "ms_per_step": round(dt / steps * 1000, 2),
         "images_per_s": round(batch * steps / dt, 1),
     }
@@ -101,6 +106,8 @@ def main():
                     help="compare eager vs hipGraph-captured step")
     ap.add_argument("--one", action="store_true",
                     help="single canonical config (for rocprof)")
+    ap.add_argument("--fused", action="store_true",
+                    help="use the kubeshare_amd fused BN+ReLU kernels")
     ap.add_argument("--batch", type=int, default=256)
     ap.add_argument("--model", default="resnet50")
     ap.add_argument("--steps", type=int, default=10)
@@ -124,7 +131,7 @@ def main():
     for batch, cl, bm, dtype, graph in cfgs:
         try:
             r = run_cfg(args.model, batch, cl, bm, dtype, args.steps,
-                        args.warmup, graph=graph)
+                        args.warmup, graph=graph, fused=args.fused)
         except RuntimeError as e:
             r = {"batch": batch, "channels_last": cl, "benchmark": bm,
                  "graph": graph, "error": str(e)[:300]}
