@@ -16,6 +16,7 @@ def main():
     ap.add_argument("--model", default="resnet50")
     ap.add_argument("--batch", type=int, default=256)
     ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--fused", action="store_true")
     args = ap.parse_args()
 
     from kubeshare_amd.utils.tuning import apply_miopen_tuning
@@ -27,8 +28,11 @@ def main():
 
     torch.backends.cudnn.benchmark = True
     dev = torch.device("cuda:0")
-    model = build_model(args.model).to(dev).to(
-        memory_format=torch.channels_last)
+    model = build_model(args.model)
+    if args.fused:
+        from kubeshare_amd import ops
+        model = ops.fuse_model(model)
+    model = model.to(dev).to(memory_format=torch.channels_last)
     opt = torch.optim.SGD(model.parameters(), lr=0.02, momentum=0.9)
     x = torch.randn(args.batch, 3, 224, 224, device=dev).contiguous(
         memory_format=torch.channels_last)
