@@ -3,19 +3,20 @@
 // Motivation (profiles/resnet50_bs256_kernel_breakdown_untuned.txt):
 // MIOpen's spatial BN kernels + separate ReLU/add elementwise passes
 // are ~48% of the non-conv GPU time of a ResNet50 training step. The
-// stock path makes 5 full passes over the activation in forward
-// (mean/var read, normalize read+write, clamp read+write) and 8 in
-// backward; these kernels make 3 and 7 (and fold the residual add and
-// its gradient for free).
+// stock path makes 5 full passes over the activation in forward and 8
+// in backward; these kernels make 3 and 7, and fold the residual add
+// and its gradient for free.
 //
-// Design (cdna_hip_programming.md G13, Appx B "Reduction"):
+// Design (cdna_hip_programming.md G13 vectorize, G7 ILP, G11 grid):
 //  - channels-last bf16, C % 8 == 0: one lane owns 8 CONSECUTIVE
-//    channels (one 16-byte load), so per-channel reductions never cross
-//    lanes — each lane accumulates f32x8 partials over its rows, a
-//    block-level LDS tree reduces the lanes that share a channel group,
-//    and one global f32 atomicAdd per channel per block merges blocks.
-//  - wave64-friendly: blockDim 256, lane -> (channel_group, row_offset)
-//    fixed for the whole kernel, so scale/bias live in registers.
+//    channels (one 16-byte load) — per-channel reductions never cross
+//    lanes, and a wave's 64 lanes cover a 1 KiB contiguous span.
+//  - UNROLL independent row-chunks per loop iteration so each lane
+//    keeps >=4 16-B loads in flight (1 load/lane measured ~2.3 TB/s,
+//    HBM-latency-bound at 32 waves/CU).
+//  - block partials + a tree-reduce in the finalize kernel instead of
+//    per-channel global atomics (2*C atomics x 2048 blocks measured as
+//    ~0.4 ms of fixed cost per BN call).
 //  - all stats/parameter math in f32; tensors read/written as bf16x8.
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
@@ -48,25 +49,39 @@ __device__ __forceinline__ unsigned short f32_to_bf16(float f) {
 }
 
 #define KS_BN_BLOCK 256
+#define KS_BN_UNROLL 4
 
 // ------------------------------------------------------------- fwd stats
-// sum[c] += x, sumsq[c] += x^2 over all M rows. Lane l owns channel
-// group cg = l % CG (8 channels), row offset l / CG.
+// Block partials: sum_part[b*C + c], sq_part[(B + b)*C + c].
 __global__ void bn_stats_kernel(const ushort8* __restrict__ x,
-                                float* __restrict__ sum,
-                                float* __restrict__ sumsq,
+                                float* __restrict__ part,
                                 long long M, int CG) {
-  __shared__ float s_sum[KS_BN_BLOCK * 8];
-  __shared__ float s_sq[KS_BN_BLOCK * 8];
+  __shared__ float s_red[KS_BN_BLOCK * 8];
   const int tid = threadIdx.x;
   const int cg = tid % CG;
   const int roff = tid / CG;
   const int rows_per_blk = KS_BN_BLOCK / CG;
+  const long long stride = (long long)gridDim.x * rows_per_blk;
 
   float8 acc = {0, 0, 0, 0, 0, 0, 0, 0};
   float8 acc2 = {0, 0, 0, 0, 0, 0, 0, 0};
   long long row = (long long)blockIdx.x * rows_per_blk + roff;
-  const long long stride = (long long)gridDim.x * rows_per_blk;
+  // unrolled main loop: KS_BN_UNROLL independent loads in flight
+  for (; row + (KS_BN_UNROLL - 1) * stride < M;
+       row += KS_BN_UNROLL * stride) {
+    ushort8 v[KS_BN_UNROLL];
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++)
+      v[u] = x[(row + u * stride) * CG + cg];
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++)
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = bf16_to_f32(v[u][j]);
+        acc[j] += f;
+        acc2[j] += f * f;
+      }
+  }
   for (; row < M; row += stride) {
     ushort8 v = x[row * CG + cg];
 #pragma unroll
@@ -76,38 +91,41 @@ __global__ void bn_stats_kernel(const ushort8* __restrict__ x,
       acc2[j] += f * f;
     }
   }
+
+  const int C = CG * 8;
+  float* sum_part = part + (long long)blockIdx.x * C;
+  float* sq_part = part + (long long)(gridDim.x + blockIdx.x) * C;
+  // LDS tree over the lanes sharing a channel group, then plain stores
 #pragma unroll
-  for (int j = 0; j < 8; j++) {
-    s_sum[tid * 8 + j] = acc[j];
-    s_sq[tid * 8 + j] = acc2[j];
-  }
+  for (int j = 0; j < 8; j++) s_red[tid * 8 + j] = acc[j];
   __syncthreads();
-  // lanes 0..CG-1 reduce over the rows_per_blk lanes sharing their cg
   if (tid < CG) {
     float8 t = {0, 0, 0, 0, 0, 0, 0, 0};
-    float8 t2 = {0, 0, 0, 0, 0, 0, 0, 0};
-    for (int r = 0; r < rows_per_blk; r++) {
-      const int src = r * CG + tid;
+    for (int r = 0; r < rows_per_blk; r++)
 #pragma unroll
-      for (int j = 0; j < 8; j++) {
-        t[j] += s_sum[src * 8 + j];
-        t2[j] += s_sq[src * 8 + j];
-      }
-    }
+      for (int j = 0; j < 8; j++) t[j] += s_red[(r * CG + tid) * 8 + j];
 #pragma unroll
-    for (int j = 0; j < 8; j++) {
-      atomicAdd(&sum[tid * 8 + j], t[j]);
-      atomicAdd(&sumsq[tid * 8 + j], t2[j]);
-    }
+    for (int j = 0; j < 8; j++) sum_part[tid * 8 + j] = t[j];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int j = 0; j < 8; j++) s_red[tid * 8 + j] = acc2[j];
+  __syncthreads();
+  if (tid < CG) {
+    float8 t = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int r = 0; r < rows_per_blk; r++)
+#pragma unroll
+      for (int j = 0; j < 8; j++) t[j] += s_red[(r * CG + tid) * 8 + j];
+#pragma unroll
+    for (int j = 0; j < 8; j++) sq_part[tid * 8 + j] = t[j];
   }
 }
 
 // ------------------------------------------------------- fwd finalize
-// mean/invstd from the reduced sums; running-stat update (PyTorch
-// semantics: running = (1-m)*running + m*batch, var unbiased); and the
-// folded per-channel scale' = w*invstd, bias' = b - mean*scale'.
-__global__ void bn_finalize_kernel(const float* __restrict__ sum,
-                                   const float* __restrict__ sumsq,
+// Reduce the block partials, then mean/invstd, running-stat update
+// (PyTorch semantics), and folded scale' = w*invstd, bias' = b - mean*s.
+__global__ void bn_finalize_kernel(const float* __restrict__ part,
+                                   int nblocks,
                                    const float* __restrict__ weight,
                                    const float* __restrict__ bias,
                                    float* __restrict__ running_mean,
@@ -120,8 +138,13 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sum,
                                    float eps) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float mean = sum[c] / (float)M;
-  float var = sumsq[c] / (float)M - mean * mean;
+  float sum = 0.f, sumsq = 0.f;
+  for (int b = 0; b < nblocks; b++) {
+    sum += part[(long long)b * C + c];
+    sumsq += part[(long long)(nblocks + b) * C + c];
+  }
+  float mean = sum / (float)M;
+  float var = sumsq / (float)M - mean * mean;
   var = var < 0.f ? 0.f : var;
   float invstd = rsqrtf(var + eps);
   save_mean[c] = mean;
@@ -137,8 +160,7 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sum,
 }
 
 // --------------------------------------------------------- fwd apply
-// y = relu(x*scale' + bias' [+ res]); scale/bias in registers (lane's
-// channel group is fixed).
+// y = relu(x*scale' + bias' [+ res]); scale/bias in registers.
 template <bool WITH_RES>
 __global__ void bn_apply_relu_kernel(const ushort8* __restrict__ x,
                                      const ushort8* __restrict__ res,
@@ -150,6 +172,7 @@ __global__ void bn_apply_relu_kernel(const ushort8* __restrict__ x,
   const int cg = tid % CG;
   const int roff = tid / CG;
   const int rows_per_blk = KS_BN_BLOCK / CG;
+  const long long stride = (long long)gridDim.x * rows_per_blk;
   float8 s, b;
 #pragma unroll
   for (int j = 0; j < 8; j++) {
@@ -157,7 +180,27 @@ __global__ void bn_apply_relu_kernel(const ushort8* __restrict__ x,
     b[j] = biasf[cg * 8 + j];
   }
   long long row = (long long)blockIdx.x * rows_per_blk + roff;
-  const long long stride = (long long)gridDim.x * rows_per_blk;
+  for (; row + (KS_BN_UNROLL - 1) * stride < M;
+       row += KS_BN_UNROLL * stride) {
+    ushort8 v[KS_BN_UNROLL], r[KS_BN_UNROLL], o[KS_BN_UNROLL];
+    long long k[KS_BN_UNROLL];
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++) {
+      k[u] = (row + u * stride) * CG + cg;
+      v[u] = x[k[u]];
+      if (WITH_RES) r[u] = res[k[u]];
+    }
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++) {
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float f = fmaf(bf16_to_f32(v[u][j]), s[j], b[j]);
+        if (WITH_RES) f += bf16_to_f32(r[u][j]);
+        o[u][j] = f32_to_bf16(f > 0.f ? f : 0.f);
+      }
+      y[k[u]] = o[u];
+    }
+  }
   for (; row < M; row += stride) {
     const long long k = row * CG + cg;
     ushort8 v = x[k];
@@ -175,9 +218,8 @@ __global__ void bn_apply_relu_kernel(const ushort8* __restrict__ x,
 }
 
 // --------------------------------------------------------- bwd stats
-// dym = dy * (y > 0); dbias[c] = sum(dym); dscale[c] = sum(dym * xhat).
-// Optionally writes dym out (it IS the residual branch's gradient for
-// the BN+Add+ReLU fusion, and the apply pass re-reads it).
+// dym = dy * (y > 0); dbias[c] = sum(dym); dscale[c] = sum(dym*xhat);
+// optionally writes dym (= the residual branch's gradient).
 template <bool WRITE_DYM>
 __global__ void bn_bwd_stats_kernel(const ushort8* __restrict__ x,
                                     const ushort8* __restrict__ y,
@@ -185,15 +227,14 @@ __global__ void bn_bwd_stats_kernel(const ushort8* __restrict__ x,
                                     ushort8* __restrict__ dym_out,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ invstd,
-                                    float* __restrict__ dbias,
-                                    float* __restrict__ dscale,
+                                    float* __restrict__ part,
                                     long long M, int CG) {
-  __shared__ float s_db[KS_BN_BLOCK * 8];
-  __shared__ float s_ds[KS_BN_BLOCK * 8];
+  __shared__ float s_red[KS_BN_BLOCK * 8];
   const int tid = threadIdx.x;
   const int cg = tid % CG;
   const int roff = tid / CG;
   const int rows_per_blk = KS_BN_BLOCK / CG;
+  const long long stride = (long long)gridDim.x * rows_per_blk;
   float8 mu, is;
 #pragma unroll
   for (int j = 0; j < 8; j++) {
@@ -203,7 +244,31 @@ __global__ void bn_bwd_stats_kernel(const ushort8* __restrict__ x,
   float8 adb = {0, 0, 0, 0, 0, 0, 0, 0};
   float8 ads = {0, 0, 0, 0, 0, 0, 0, 0};
   long long row = (long long)blockIdx.x * rows_per_blk + roff;
-  const long long stride = (long long)gridDim.x * rows_per_blk;
+  for (; row + (KS_BN_UNROLL - 1) * stride < M;
+       row += KS_BN_UNROLL * stride) {
+    ushort8 xv[KS_BN_UNROLL], yv[KS_BN_UNROLL], gv[KS_BN_UNROLL];
+    long long k[KS_BN_UNROLL];
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++) {
+      k[u] = (row + u * stride) * CG + cg;
+      xv[u] = x[k[u]];
+      yv[u] = y[k[u]];
+      gv[u] = dy[k[u]];
+    }
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++) {
+      ushort8 dm;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float g = bf16_to_f32(yv[u][j]) > 0.f ? bf16_to_f32(gv[u][j]) : 0.f;
+        float xhat = (bf16_to_f32(xv[u][j]) - mu[j]) * is[j];
+        adb[j] += g;
+        ads[j] += g * xhat;
+        if (WRITE_DYM) dm[j] = f32_to_bf16(g);
+      }
+      if (WRITE_DYM) dym_out[k[u]] = dm;
+    }
+  }
   for (; row < M; row += stride) {
     const long long k = row * CG + cg;
     ushort8 xv = x[k], yv = y[k], gv = dy[k];
@@ -218,29 +283,49 @@ __global__ void bn_bwd_stats_kernel(const ushort8* __restrict__ x,
     }
     if (WRITE_DYM) dym_out[k] = dm;
   }
+
+  const int C = CG * 8;
+  float* db_part = part + (long long)blockIdx.x * C;
+  float* ds_part = part + (long long)(gridDim.x + blockIdx.x) * C;
 #pragma unroll
-  for (int j = 0; j < 8; j++) {
-    s_db[tid * 8 + j] = adb[j];
-    s_ds[tid * 8 + j] = ads[j];
-  }
+  for (int j = 0; j < 8; j++) s_red[tid * 8 + j] = adb[j];
   __syncthreads();
   if (tid < CG) {
     float8 t = {0, 0, 0, 0, 0, 0, 0, 0};
-    float8 t2 = {0, 0, 0, 0, 0, 0, 0, 0};
-    for (int r = 0; r < rows_per_blk; r++) {
-      const int src = r * CG + tid;
+    for (int r = 0; r < rows_per_blk; r++)
 #pragma unroll
-      for (int j = 0; j < 8; j++) {
-        t[j] += s_db[src * 8 + j];
-        t2[j] += s_ds[src * 8 + j];
-      }
-    }
+      for (int j = 0; j < 8; j++) t[j] += s_red[(r * CG + tid) * 8 + j];
 #pragma unroll
-    for (int j = 0; j < 8; j++) {
-      atomicAdd(&dbias[tid * 8 + j], t[j]);
-      atomicAdd(&dscale[tid * 8 + j], t2[j]);
-    }
+    for (int j = 0; j < 8; j++) db_part[tid * 8 + j] = t[j];
   }
+  __syncthreads();
+#pragma unroll
+  for (int j = 0; j < 8; j++) s_red[tid * 8 + j] = ads[j];
+  __syncthreads();
+  if (tid < CG) {
+    float8 t = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int r = 0; r < rows_per_blk; r++)
+#pragma unroll
+      for (int j = 0; j < 8; j++) t[j] += s_red[(r * CG + tid) * 8 + j];
+#pragma unroll
+    for (int j = 0; j < 8; j++) ds_part[tid * 8 + j] = t[j];
+  }
+}
+
+// reduce bwd partials -> dbias[c], dscale[c]
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ part,
+                                       int nblocks,
+                                       float* __restrict__ dbias,
+                                       float* __restrict__ dscale, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float db = 0.f, ds = 0.f;
+  for (int b = 0; b < nblocks; b++) {
+    db += part[(long long)b * C + c];
+    ds += part[(long long)(nblocks + b) * C + c];
+  }
+  dbias[c] = db;
+  dscale[c] = ds;
 }
 
 // --------------------------------------------------------- bwd apply
@@ -259,6 +344,7 @@ __global__ void bn_bwd_apply_kernel(const ushort8* __restrict__ x,
   const int cg = tid % CG;
   const int roff = tid / CG;
   const int rows_per_blk = KS_BN_BLOCK / CG;
+  const long long stride = (long long)gridDim.x * rows_per_blk;
   const float invM = 1.f / (float)M;
   float8 mu, is, w, db, ds;
 #pragma unroll
@@ -271,7 +357,29 @@ __global__ void bn_bwd_apply_kernel(const ushort8* __restrict__ x,
     ds[j] = dscale[c] * invM;
   }
   long long row = (long long)blockIdx.x * rows_per_blk + roff;
-  const long long stride = (long long)gridDim.x * rows_per_blk;
+  for (; row + (KS_BN_UNROLL - 1) * stride < M;
+       row += KS_BN_UNROLL * stride) {
+    ushort8 xv[KS_BN_UNROLL], yv[KS_BN_UNROLL], gv[KS_BN_UNROLL];
+    long long k[KS_BN_UNROLL];
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++) {
+      k[u] = (row + u * stride) * CG + cg;
+      xv[u] = x[k[u]];
+      yv[u] = y[k[u]];
+      gv[u] = dy[k[u]];
+    }
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++) {
+      ushort8 o;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float g = bf16_to_f32(yv[u][j]) > 0.f ? bf16_to_f32(gv[u][j]) : 0.f;
+        float xhat = (bf16_to_f32(xv[u][j]) - mu[j]) * is[j];
+        o[j] = f32_to_bf16(w[j] * (g - db[j] - xhat * ds[j]));
+      }
+      dx[k[u]] = o;
+    }
+  }
   for (; row < M; row += stride) {
     const long long k = row * CG + cg;
     ushort8 xv = x[k], yv = y[k], gv = dy[k];
@@ -306,7 +414,7 @@ namespace {
 
 struct Geom {
   long long M;
-  int C, CG, blocks;
+  int C, CG, blocks, stat_blocks;
 };
 
 Geom geom_of(const torch::Tensor& x) {
@@ -324,6 +432,11 @@ Geom geom_of(const torch::Tensor& x) {
   long long blocks = (work + KS_BN_BLOCK - 1) / KS_BN_BLOCK;
   // G11: cap + grid-stride (256 CUs want >>256 workgroups)
   g.blocks = (int)std::min<long long>(blocks, 2048);
+  // stats kernels pay nblocks*C partial traffic in finalize: fewer,
+  // fatter blocks (the unrolled loop keeps them latency-covered)
+  g.stat_blocks = (int)std::min<long long>(
+      (blocks + KS_BN_UNROLL - 1) / KS_BN_UNROLL, 1024);
+  if (g.stat_blocks < 1) g.stat_blocks = 1;
   return g;
 }
 
@@ -336,22 +449,20 @@ std::vector<torch::Tensor> bn_relu_fwd_train(
   Geom g = geom_of(x);
   auto stream = at::cuda::getCurrentCUDAStream();
   auto f32 = x.options().dtype(at::kFloat);
-  auto sum = at::zeros({g.C * 2}, f32);
-  float* sum_p = sum.data_ptr<float>();
-  float* sumsq_p = sum_p + g.C;
+  auto part = at::empty({(long long)2 * g.stat_blocks * g.C}, f32);
   auto y = at::empty_like(x);
   auto save_mean = at::empty({g.C}, f32);
   auto save_invstd = at::empty({g.C}, f32);
   auto scale = at::empty({g.C}, f32);
   auto biasf = at::empty({g.C}, f32);
 
-  hipLaunchKernelGGL(bn_stats_kernel, dim3(g.blocks), dim3(KS_BN_BLOCK), 0,
-                     stream.stream(),
-                     reinterpret_cast<const ushort8*>(x.data_ptr()), sum_p,
-                     sumsq_p, g.M, g.CG);
+  hipLaunchKernelGGL(bn_stats_kernel, dim3(g.stat_blocks),
+                     dim3(KS_BN_BLOCK), 0, stream.stream(),
+                     reinterpret_cast<const ushort8*>(x.data_ptr()),
+                     part.data_ptr<float>(), g.M, g.CG);
   int fb = (g.C + 255) / 256;
   hipLaunchKernelGGL(bn_finalize_kernel, dim3(fb), dim3(256), 0,
-                     stream.stream(), sum_p, sumsq_p,
+                     stream.stream(), part.data_ptr<float>(), g.stat_blocks,
                      weight.data_ptr<float>(), bias.data_ptr<float>(),
                      running_mean.defined()
                          ? running_mean.data_ptr<float>() : nullptr,
@@ -422,7 +533,8 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor x, torch::Tensor y,
   Geom g = geom_of(x);
   auto stream = at::cuda::getCurrentCUDAStream();
   auto f32 = x.options().dtype(at::kFloat);
-  auto red = at::zeros({g.C * 2}, f32);
+  auto part = at::empty({(long long)2 * g.stat_blocks * g.C}, f32);
+  auto red = at::empty({g.C * 2}, f32);
   float* dbias_p = red.data_ptr<float>();
   float* dscale_p = dbias_p + g.C;
   auto dx = at::empty_like(x);
@@ -432,24 +544,28 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor x, torch::Tensor y,
 
   if (need_dres) {
     dres = at::empty_like(x);
-    hipLaunchKernelGGL(bn_bwd_stats_kernel<true>, dim3(g.blocks),
+    hipLaunchKernelGGL(bn_bwd_stats_kernel<true>, dim3(g.stat_blocks),
                        dim3(KS_BN_BLOCK), 0, stream.stream(),
                        reinterpret_cast<const ushort8*>(x.data_ptr()),
                        reinterpret_cast<const ushort8*>(y.data_ptr()),
                        reinterpret_cast<const ushort8*>(dy.data_ptr()),
                        reinterpret_cast<ushort8*>(dres.data_ptr()),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       dbias_p, dscale_p, g.M, g.CG);
+                       part.data_ptr<float>(), g.M, g.CG);
   } else {
-    hipLaunchKernelGGL(bn_bwd_stats_kernel<false>, dim3(g.blocks),
+    hipLaunchKernelGGL(bn_bwd_stats_kernel<false>, dim3(g.stat_blocks),
                        dim3(KS_BN_BLOCK), 0, stream.stream(),
                        reinterpret_cast<const ushort8*>(x.data_ptr()),
                        reinterpret_cast<const ushort8*>(y.data_ptr()),
                        reinterpret_cast<const ushort8*>(dy.data_ptr()),
                        nullptr, mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), dbias_p, dscale_p, g.M,
-                       g.CG);
+                       invstd.data_ptr<float>(), part.data_ptr<float>(),
+                       g.M, g.CG);
   }
+  int fb = (g.C + 255) / 256;
+  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(fb), dim3(256), 0,
+                     stream.stream(), part.data_ptr<float>(), g.stat_blocks,
+                     dbias_p, dscale_p, g.C);
   hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(g.blocks), dim3(KS_BN_BLOCK),
                      0, stream.stream(),
                      reinterpret_cast<const ushort8*>(x.data_ptr()),
