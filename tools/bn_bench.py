@@ -73,6 +73,33 @@ def main():
             xg.grad = None
             resg.grad = None
 
+        # GPU-only throughput: launch L iterations back-to-back, time
+        # with events (host launch gaps overlap, so this is kernel time)
+        ext = ops._load()
+        w32 = bn2.weight.detach().float()
+        b32 = bn2.bias.detach().float()
+
+        def gpu_only(fn, iters=30):
+            fn()
+            sync()
+            ev0 = torch.cuda.Event(enable_timing=True)
+            ev1 = torch.cuda.Event(enable_timing=True)
+            ev0.record()
+            for _ in range(iters):
+                fn()
+            ev1.record()
+            sync()
+            return ev0.elapsed_time(ev1) / iters
+
+        def ext_fwd():
+            ext.bn_relu_fwd_train(x, w32, b32, bn2.running_mean,
+                                  bn2.running_var, 0.1, 1e-5, res)
+
+        def aten_fwd():
+            torch.ops.aten.miopen_batch_norm(
+                x.float(), bn1.weight, bn1.bias, bn1.running_mean,
+                bn1.running_var, True, 0.1, 1e-5)
+
         r = {
             "shape": [n, c, h, w],
             "mb": round(x.numel() * 2 / 1e6, 1),
@@ -80,6 +107,8 @@ def main():
             "fused_fwd_ms": round(bench(fused_fwd, sync), 3),
             "stock_fb_ms": round(bench(stock_fb, sync), 3),
             "fused_fb_ms": round(bench(fused_fb, sync), 3),
+            "gpuonly_ext_fwd_ms": round(gpu_only(ext_fwd), 3),
+            "gpuonly_aten_fwd_ms": round(gpu_only(aten_fwd), 3),
         }
         r["fwd_speedup"] = round(r["stock_fwd_ms"] / r["fused_fwd_ms"], 2)
         r["fb_speedup"] = round(r["stock_fb_ms"] / r["fused_fb_ms"], 2)
