@@ -50,6 +50,7 @@ __device__ __forceinline__ unsigned short f32_to_bf16(float f) {
 
 #define KS_BN_BLOCK 256
 #define KS_BN_UNROLL 4
+#define KS_BN_UNROLL_STATS 8
 
 // ------------------------------------------------------------- fwd stats
 // Block partials: sum_part[b*C + c], sq_part[(B + b)*C + c].
@@ -66,15 +67,15 @@ __global__ void bn_stats_kernel(const ushort8* __restrict__ x,
   float8 acc = {0, 0, 0, 0, 0, 0, 0, 0};
   float8 acc2 = {0, 0, 0, 0, 0, 0, 0, 0};
   long long row = (long long)blockIdx.x * rows_per_blk + roff;
-  // unrolled main loop: KS_BN_UNROLL independent loads in flight
-  for (; row + (KS_BN_UNROLL - 1) * stride < M;
-       row += KS_BN_UNROLL * stride) {
-    ushort8 v[KS_BN_UNROLL];
+  // unrolled main loop: KS_BN_UNROLL_STATS independent loads in flight
+  for (; row + (KS_BN_UNROLL_STATS - 1) * stride < M;
+       row += KS_BN_UNROLL_STATS * stride) {
+    ushort8 v[KS_BN_UNROLL_STATS];
 #pragma unroll
-    for (int u = 0; u < KS_BN_UNROLL; u++)
+    for (int u = 0; u < KS_BN_UNROLL_STATS; u++)
       v[u] = x[(row + u * stride) * CG + cg];
 #pragma unroll
-    for (int u = 0; u < KS_BN_UNROLL; u++)
+    for (int u = 0; u < KS_BN_UNROLL_STATS; u++)
 #pragma unroll
       for (int j = 0; j < 8; j++) {
         float f = bf16_to_f32(v[u][j]);
@@ -121,11 +122,38 @@ __global__ void bn_stats_kernel(const ushort8* __restrict__ x,
   }
 }
 
+// ------------------------------------------------- partial reduction
+// part[b*C + c] (+ the sq/dscale half at offset nblocks*C) -> out[c].
+// One 64-lane wave per 4 channels: lanes stride over blocks (all loads
+// independent -> one latency, not nblocks of them), then shfl-reduce.
+__global__ void bn_reduce_partials_kernel(const float* __restrict__ part,
+                                          int nblocks, int C,
+                                          float* __restrict__ out0,
+                                          float* __restrict__ out1) {
+  const int lane = threadIdx.x;        // 0..63 over blocks
+  const int c = blockIdx.x * blockDim.y + threadIdx.y;
+  if (c >= C) return;
+  float a0 = 0.f, a1 = 0.f;
+  for (int b = lane; b < nblocks; b += 64) {
+    a0 += part[(long long)b * C + c];
+    a1 += part[(long long)(nblocks + b) * C + c];
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    a0 += __shfl_down(a0, off, 64);
+    a1 += __shfl_down(a1, off, 64);
+  }
+  if (lane == 0) {
+    out0[c] = a0;
+    out1[c] = a1;
+  }
+}
+
 // ------------------------------------------------------- fwd finalize
 // Reduce the block partials, then mean/invstd, running-stat update
 // (PyTorch semantics), and folded scale' = w*invstd, bias' = b - mean*s.
-__global__ void bn_finalize_kernel(const float* __restrict__ part,
-                                   int nblocks,
+__global__ void bn_finalize_kernel(const float* __restrict__ sum,
+                                   const float* __restrict__ sumsq,
                                    const float* __restrict__ weight,
                                    const float* __restrict__ bias,
                                    float* __restrict__ running_mean,
@@ -138,13 +166,8 @@ __global__ void bn_finalize_kernel(const float* __restrict__ part,
                                    float eps) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float sum = 0.f, sumsq = 0.f;
-  for (int b = 0; b < nblocks; b++) {
-    sum += part[(long long)b * C + c];
-    sumsq += part[(long long)(nblocks + b) * C + c];
-  }
-  float mean = sum / (float)M;
-  float var = sumsq / (float)M - mean * mean;
+  float mean = sum[c] / (float)M;
+  float var = sumsq[c] / (float)M - mean * mean;
   var = var < 0.f ? 0.f : var;
   float invstd = rsqrtf(var + eps);
   save_mean[c] = mean;
@@ -312,22 +335,6 @@ __global__ void bn_bwd_stats_kernel(const ushort8* __restrict__ x,
   }
 }
 
-// reduce bwd partials -> dbias[c], dscale[c]
-__global__ void bn_bwd_finalize_kernel(const float* __restrict__ part,
-                                       int nblocks,
-                                       float* __restrict__ dbias,
-                                       float* __restrict__ dscale, int C) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float db = 0.f, ds = 0.f;
-  for (int b = 0; b < nblocks; b++) {
-    db += part[(long long)b * C + c];
-    ds += part[(long long)(nblocks + b) * C + c];
-  }
-  dbias[c] = db;
-  dscale[c] = ds;
-}
-
 // --------------------------------------------------------- bwd apply
 // dx = w*invstd * (dym - dbias/M - xhat * dscale/M)
 __global__ void bn_bwd_apply_kernel(const ushort8* __restrict__ x,
@@ -432,10 +439,7 @@ Geom geom_of(const torch::Tensor& x) {
   long long blocks = (work + KS_BN_BLOCK - 1) / KS_BN_BLOCK;
   // G11: cap + grid-stride (256 CUs want >>256 workgroups)
   g.blocks = (int)std::min<long long>(blocks, 2048);
-  // stats kernels pay nblocks*C partial traffic in finalize: fewer,
-  // fatter blocks (the unrolled loop keeps them latency-covered)
-  g.stat_blocks = (int)std::min<long long>(
-      (blocks + KS_BN_UNROLL - 1) / KS_BN_UNROLL, 1024);
+  g.stat_blocks = (int)std::min<long long>(blocks, 1024);
   if (g.stat_blocks < 1) g.stat_blocks = 1;
   return g;
 }
@@ -455,14 +459,20 @@ std::vector<torch::Tensor> bn_relu_fwd_train(
   auto save_invstd = at::empty({g.C}, f32);
   auto scale = at::empty({g.C}, f32);
   auto biasf = at::empty({g.C}, f32);
+  auto sums = at::empty({2 * g.C}, f32);
 
   hipLaunchKernelGGL(bn_stats_kernel, dim3(g.stat_blocks),
                      dim3(KS_BN_BLOCK), 0, stream.stream(),
                      reinterpret_cast<const ushort8*>(x.data_ptr()),
                      part.data_ptr<float>(), g.M, g.CG);
+  hipLaunchKernelGGL(bn_reduce_partials_kernel,
+                     dim3((g.C + 3) / 4), dim3(64, 4), 0, stream.stream(),
+                     part.data_ptr<float>(), g.stat_blocks, g.C,
+                     sums.data_ptr<float>(), sums.data_ptr<float>() + g.C);
   int fb = (g.C + 255) / 256;
   hipLaunchKernelGGL(bn_finalize_kernel, dim3(fb), dim3(256), 0,
-                     stream.stream(), part.data_ptr<float>(), g.stat_blocks,
+                     stream.stream(), sums.data_ptr<float>(),
+                     sums.data_ptr<float>() + g.C,
                      weight.data_ptr<float>(), bias.data_ptr<float>(),
                      running_mean.defined()
                          ? running_mean.data_ptr<float>() : nullptr,
@@ -562,10 +572,10 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor x, torch::Tensor y,
                        invstd.data_ptr<float>(), part.data_ptr<float>(),
                        g.M, g.CG);
   }
-  int fb = (g.C + 255) / 256;
-  hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(fb), dim3(256), 0,
-                     stream.stream(), part.data_ptr<float>(), g.stat_blocks,
-                     dbias_p, dscale_p, g.C);
+  hipLaunchKernelGGL(bn_reduce_partials_kernel,
+                     dim3((g.C + 3) / 4), dim3(64, 4), 0, stream.stream(),
+                     part.data_ptr<float>(), g.stat_blocks, g.C,
+                     dbias_p, dscale_p);
   hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(g.blocks), dim3(KS_BN_BLOCK),
                      0, stream.stream(),
                      reinterpret_cast<const ushort8*>(x.data_ptr()),
