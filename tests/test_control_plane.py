@@ -193,3 +193,39 @@ def test_full_stack_e2e(tmp_path, native_bins, fake_gpus):
             assert st["pods"][p.key]["grants"] >= 1
     finally:
         nd.stop()
+
+
+def test_simulator_smoke():
+    """Trace-driven scheduler load test (reference test/simulator) on
+    the in-memory harness: 150 jobs, bounded cycle time, no wedge."""
+    import subprocess
+    import sys as _sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [_sys.executable, "tools/simulator.py", "--jobs", "150",
+         "--nodes", "2"],
+        cwd=repo, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-1500:]
+    out = json.loads(r.stdout.strip().splitlines()[-1])
+    assert out["jobs"] == 150
+    assert out["p99_cycle_ms"] < 50.0
+
+
+def test_sharepod_conversion():
+    from kubeshare_amd.sharepod import sharepod_to_pod
+    obj = {
+        "apiVersion": "sharedgpu.kubeshare.amd/v1", "kind": "SharePod",
+        "metadata": {"name": "sp1", "namespace": "team", "uid": "u-1"},
+        "spec": {
+            "gpuRequest": "0.5", "gpuLimit": "1.0", "priority": "100",
+            "groupName": "g", "groupHeadcount": "2", "groupThreshold": "1.0",
+            "template": {"spec": {"containers": [
+                {"name": "main", "image": "rocm/pytorch"}]}},
+        },
+    }
+    pod = sharepod_to_pod(obj)
+    assert pod["metadata"]["name"] == "sp1"
+    assert pod["metadata"]["labels"][C.POD_GPU_REQUEST] == "0.5"
+    assert pod["metadata"]["labels"][C.POD_GROUP_NAME] == "g"
+    assert pod["spec"]["schedulerName"] == "kubeshare-scheduler"
+    assert pod["metadata"]["ownerReferences"][0]["uid"] == "u-1"
