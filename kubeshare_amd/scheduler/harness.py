@@ -11,7 +11,7 @@ from dataclasses import dataclass, field
 
 from ..utils import constants as C
 from .inventory import FakeInventory
-from .plugin import KubeShareScheduler, Placement, QueuedPodInfo
+from .plugin import KubeShareScheduler, Placement
 from .topology import TopologyConfig
 
 
@@ -86,16 +86,16 @@ class FakeCluster:
         return self
 
     def _queue(self):
+        """Reference QueueSort order (scheduler.go:247-267): group
+        priority desc, then enqueue time, then key. All pending pods
+        share this simulated cycle's clock, so time degenerates to
+        the key tiebreak."""
         pending = [p for p in self.pods.values() if p.phase == "Pending"]
-        infos = {p.key: QueuedPodInfo(
-            spec=self.scheduler.pod_status.get(p.key), timestamp=self.clock)
-            for p in pending}
 
         def sort_key(p):
             spec = self.scheduler.pod_status.get(p.key)
             prio = spec.priority if spec else 0
             return (-prio, p.key)
-        del infos
         return sorted(pending, key=sort_key)
 
     def schedule_one(self, pod: FakePod) -> str:

@@ -1,0 +1,106 @@
+"""Unit tests for the smaller scheduler pieces: QueueSort ordering,
+score normalization, port pool, topology helpers, models."""
+import pytest
+
+from kubeshare_amd.scheduler.bitmap import RRPortPool
+from kubeshare_amd.scheduler.harness import FakeCluster
+from kubeshare_amd.scheduler.inventory import FakeInventory
+from kubeshare_amd.scheduler.plugin import KubeShareScheduler, QueuedPodInfo
+from kubeshare_amd.scheduler.topology import TopologyConfig
+from kubeshare_amd.utils import constants as C
+from kubeshare_amd.utils.labels import PodSpec
+
+
+def _sched():
+    return KubeShareScheduler(TopologyConfig.single_node("n"))
+
+
+def _info(name, prio, ts):
+    return QueuedPodInfo(spec=PodSpec("ns", name, priority=prio),
+                        timestamp=ts)
+
+
+def test_queue_sort_less():
+    """Reference Less (scheduler.go:247-267): priority desc, then
+    timestamp, then key."""
+    s = _sched()
+    assert s.less(_info("a", 100, 5.0), _info("b", 0, 1.0))
+    assert not s.less(_info("a", 0, 5.0), _info("b", 100, 1.0))
+    assert s.less(_info("a", 50, 1.0), _info("b", 50, 2.0))
+    assert s.less(_info("a", 50, 1.0), _info("b", 50, 1.0))  # key tiebreak
+
+
+def test_normalize_scores():
+    """Reference NormalizeScore (scheduler.go:443-487): shift negatives,
+    rescale to 0..100."""
+    s = KubeShareScheduler.normalize_scores(
+        {"a": -50.0, "b": 0.0, "c": 50.0})
+    assert s["a"] == 0.0
+    assert s["c"] == 100.0
+    assert 0.0 < s["b"] < 100.0
+    assert KubeShareScheduler.normalize_scores({"a": 0.0}) == {"a": 0.0}
+    assert KubeShareScheduler.normalize_scores({}) == {}
+
+
+def test_port_pool_round_robin():
+    pool = RRPortPool(base=50050, size=4)
+    p1, p2 = pool.allocate(), pool.allocate()
+    assert (p1, p2) == (50050, 50051)
+    pool.release(p1)
+    # round-robin: the just-released port is NOT immediately reused
+    assert pool.allocate() == 50052
+    assert pool.allocate() == 50053
+    assert pool.allocate() == 50050  # wraps to the released one
+    assert not pool.available()
+    with pytest.raises(RuntimeError):
+        pool.allocate()
+
+
+def test_port_pool_mark_for_resync():
+    pool = RRPortPool(base=50050, size=4)
+    pool.mark(50051)
+    got = {pool.allocate() for _ in range(3)}
+    assert 50051 not in got
+
+
+def test_filter_rejects_when_ports_exhausted():
+    fc = FakeCluster(nodes={"node-a": {"gpus": 1}})
+    sch = fc.scheduler
+    pool = sch.ports["node-a"]
+    for _ in range(pool.size):
+        pool.allocate()
+    pod = fc.add_pod("ns", "p", {C.POD_GPU_REQUEST: "0.5",
+                                 C.POD_GPU_LIMIT: "1.0"})
+    fc.schedule_pending()
+    assert pod.phase == "Unschedulable"
+
+
+def test_single_node_topology_helper():
+    cfg = TopologyConfig.single_node("host1", gpus=4)
+    s = KubeShareScheduler(cfg)
+    inv = FakeInventory({"host1": {"gpus": 4}})
+    s.register_node("host1", inv.by_model("host1"))
+    assert len(s.tree.leaves_on_node("host1")) == 4
+
+
+def test_fake_inventory_xgmi_clique():
+    inv = FakeInventory({"n": {"gpus": 8}})
+    gpus = inv.gpus("n")
+    for g in gpus:
+        assert len(g.xgmi_links) == 7  # MI355X: 7 p2p links per GPU
+        assert g.index not in g.xgmi_links
+
+
+def test_vgg16_forward_cpu():
+    import torch
+    from kubeshare_amd.models import build_model
+    net = build_model("vgg16", num_classes=10)
+    out = net(torch.randn(1, 3, 64, 64))
+    assert out.shape == (1, 10)
+
+
+def test_resnet18_forward_cpu():
+    import torch
+    from kubeshare_amd.models import build_model
+    net = build_model("resnet18", num_classes=7)
+    assert net(torch.randn(2, 3, 64, 64)).shape == (2, 7)

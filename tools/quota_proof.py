@@ -15,7 +15,6 @@ import json
 import os
 import subprocess
 import sys
-import time
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, REPO)
@@ -60,14 +59,12 @@ def main():
         for p in (pa, pb):
             line = p.stdout.readline().strip()
             assert line == "READY", line
-        t0 = time.time()
         for p in (pa, pb):
             p.stdin.write("GO\n")
             p.stdin.flush()
         out_a, _ = pa.communicate(timeout=args.duration_ms / 1000 + 120)
         out_b, _ = pb.communicate(timeout=120)
         wall_s = args.duration_ms / 1000.0
-        del t0
         st = share.stats()
     finally:
         share.stop()
