@@ -72,6 +72,7 @@ def main():
     ap.add_argument("--device", default=None,
                     help="override (cpu for plumbing tests)")
     ap.add_argument("--use-ops", default="auto")
+    ap.add_argument("--fused-sgd", action="store_true")
     args = ap.parse_args()
 
     import torch
@@ -119,7 +120,8 @@ def main():
                    "--steps", str(args.steps), "--warmup", str(args.warmup),
                    "--device", wdev, "--dtype",
                    "bf16" if args.dtype == "bf16" else "fp32",
-                   "--use-ops", args.use_ops]
+                   "--use-ops", args.use_ops] \
+                  + (["--fused-sgd"] if args.fused_sgd else [])
             env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
             p = subprocess.Popen(cmd, env=env, cwd=REPO,
                                  stdin=subprocess.PIPE,

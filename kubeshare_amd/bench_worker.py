@@ -54,6 +54,8 @@ def main():
     ap.add_argument("--lr", type=float, default=0.02)
     ap.add_argument("--use-ops", default="auto", choices=["auto", "on", "off"],
                     help="use the kubeshare_amd HIP fused ops")
+    ap.add_argument("--fused-sgd", action="store_true",
+                    help="use the HIP multi-tensor SGD instead of torch's")
     args = ap.parse_args()
 
     from kubeshare_amd.utils.tuning import apply_miopen_tuning
@@ -85,8 +87,13 @@ def main():
     if channels_last:
         model = model.to(memory_format=torch.channels_last)
 
-    opt = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.9,
-                          weight_decay=1e-4)
+    if use_ops and args.fused_sgd:
+        from kubeshare_amd import ops as _ops
+        opt = _ops.FusedSGD(model.parameters(), lr=args.lr, momentum=0.9,
+                            weight_decay=1e-4)
+    else:
+        opt = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.9,
+                              weight_decay=1e-4)
     loss_fn = torch.nn.CrossEntropyLoss()
 
     # synthetic data, fixed on-device batch (BASELINE: no network for

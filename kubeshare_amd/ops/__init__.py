@@ -81,6 +81,8 @@ class FusedSGD:
         self.lr = lr
         self.momentum = momentum
         self.weight_decay = weight_decay
+        # momentum buffers share the parameter's layout (channels-last
+        # conv weights included): the kernel updates flat dense storage
         self.momenta = [torch.zeros_like(p) for p in self.params]
         self._first = True
 

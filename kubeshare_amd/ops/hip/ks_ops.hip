@@ -100,9 +100,14 @@ void sgd_momentum_(std::vector<torch::Tensor> params,
     auto& g = grads[t];
     auto& v = momenta[t];
     TORCH_CHECK(p.is_cuda() && p.scalar_type() == at::kFloat &&
-                    p.is_contiguous() && g.is_contiguous() &&
-                    v.is_contiguous(),
-                "sgd_momentum_: f32 contiguous only");
+                    p.is_non_overlapping_and_dense() &&
+                    g.sizes() == p.sizes() && g.strides() == p.strides() &&
+                    v.strides() == p.strides() &&
+                    g.scalar_type() == at::kFloat &&
+                    v.scalar_type() == at::kFloat,
+                "sgd_momentum_: f32 dense tensors with matching strides "
+                "(any memory format: the update is elementwise over the "
+                "flat storage)");
     long long n = p.numel();
     int threads = 256;
     int blocks = (int)std::min<long long>(2048, (n / 4 + threads - 1) / threads + 1);

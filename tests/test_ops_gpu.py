@@ -82,6 +82,30 @@ def test_bn_relu_forward_backward(shape, with_res):
                                    rtol=5e-2, atol=5e-2)
 
 
+def test_sgd_momentum_channels_last():
+    """FusedSGD on channels-last conv weights (the flagship layout):
+    matches torch.optim.SGD."""
+    torch.manual_seed(3)
+    conv = torch.nn.Conv2d(64, 128, 3, bias=False).cuda().to(
+        memory_format=torch.channels_last)
+    ref = conv.weight.detach().clone()
+    g = torch.randn_like(conv.weight)
+
+    ref_p = ref.clone().requires_grad_()
+    ref_p.grad = g.clone()
+    opt = torch.optim.SGD([ref_p], lr=0.1, momentum=0.9, weight_decay=1e-4)
+    for _ in range(3):
+        opt.step()
+
+    fused = ops.FusedSGD([conv.weight.requires_grad_()], lr=0.1,
+                         momentum=0.9, weight_decay=1e-4)
+    conv.weight.grad = g.clone()
+    for _ in range(3):
+        fused.step()
+    torch.cuda.synchronize()
+    torch.testing.assert_close(conv.weight, ref_p, rtol=1e-5, atol=1e-6)
+
+
 def test_bn_relu_eval_mode():
     x = _mk((4, 64, 14, 14))
     bn = torch.nn.BatchNorm2d(64).cuda()
