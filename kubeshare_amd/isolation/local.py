@@ -102,17 +102,30 @@ class LocalGPUShare:
         os.makedirs(self.config_dir, exist_ok=True)
         os.makedirs(self.port_dir, exist_ok=True)
         F.write_gpu_config(self.config_dir, self.uuid, [])
-        if not self.sched_port:
-            self.sched_port = free_port()
         log = os.path.join(self.workdir, f"gpu-schd-{self.gpu_index}.log")
-        self._schd = subprocess.Popen(
-            [native_path("gpu-schd"), "-p", self.config_dir, "-f", self.uuid,
-             "-P", str(self.sched_port), "-q", str(self.base_quota_ms),
-             "-m", str(self.min_quota_ms), "-w", str(self.window_ms),
-             "-l", log],
-            stderr=subprocess.DEVNULL)
-        _wait_port(self.sched_port)
-        return self
+        # free_port() is bind-then-release: with 8 ranks starting their
+        # stacks concurrently (the N=8 scaling bench) another process
+        # can grab the port in between — retry with a fresh one
+        last_err = None
+        for _ in range(4):
+            port = self.sched_port or free_port()
+            self._schd = subprocess.Popen(
+                [native_path("gpu-schd"), "-p", self.config_dir,
+                 "-f", self.uuid, "-P", str(port),
+                 "-q", str(self.base_quota_ms),
+                 "-m", str(self.min_quota_ms), "-w", str(self.window_ms),
+                 "-l", log],
+                stderr=subprocess.DEVNULL)
+            try:
+                _wait_port(port, timeout=10.0)
+                self.sched_port = port
+                return self
+            except TimeoutError as e:
+                last_err = e
+                self._schd.kill()
+                self._schd.wait()
+                self.sched_port = 0
+        raise last_err
 
     def add_pod(self, name: str, request: float, limit: float | None = None,
                 memory: int = 0) -> PodHandle:
@@ -120,22 +133,32 @@ class LocalGPUShare:
         if memory <= 0:
             memory = math.floor(request * self.full_memory)
         h = PodHandle(name=name, request=request, limit=limit, memory=memory,
-                      manager_port=free_port())
-        self.pods[name] = h
-        self._rewrite_config()
-        env = dict(os.environ)
-        env.update({
-            C.ENV_SCHEDULER_IP: "127.0.0.1",
-            C.ENV_SCHEDULER_PORT: str(self.sched_port),
-            C.ENV_POD_MANAGER_IP: "0.0.0.0",
-            C.ENV_POD_MANAGER_PORT: str(h.manager_port),
-            C.ENV_POD_NAME: name,
-            "POD_MANAGER_LOG": os.path.join(self.workdir, "pod-mgr.log"),
-        })
-        h.manager_proc = subprocess.Popen([native_path("pod-mgr")], env=env,
-                                          stderr=subprocess.DEVNULL)
-        _wait_port(h.manager_port)
-        return h
+                      manager_port=0)
+        last_err = None
+        for _ in range(4):  # see start(): concurrent-rank port races
+            h.manager_port = free_port()
+            env = dict(os.environ)
+            env.update({
+                C.ENV_SCHEDULER_IP: "127.0.0.1",
+                C.ENV_SCHEDULER_PORT: str(self.sched_port),
+                C.ENV_POD_MANAGER_IP: "0.0.0.0",
+                C.ENV_POD_MANAGER_PORT: str(h.manager_port),
+                C.ENV_POD_NAME: name,
+                "POD_MANAGER_LOG": os.path.join(self.workdir, "pod-mgr.log"),
+            })
+            h.manager_proc = subprocess.Popen([native_path("pod-mgr")],
+                                              env=env,
+                                              stderr=subprocess.DEVNULL)
+            try:
+                _wait_port(h.manager_port, timeout=10.0)
+                self.pods[name] = h
+                self._rewrite_config()
+                return h
+            except TimeoutError as e:
+                last_err = e
+                h.manager_proc.kill()
+                h.manager_proc.wait()
+        raise last_err
 
     def remove_pod(self, name: str):
         h = self.pods.pop(name, None)
