@@ -167,6 +167,13 @@ class NodeDaemon:
                 continue
             sup.reconcile(entries)
 
+    def serve_metrics(self, port: int, node_name: str = ""):
+        """Export every local gpu-schd's STATS as Prometheus metrics."""
+        from .metrics import serve
+        endpoints = {uuid: ("127.0.0.1", sup.sched_port)
+                     for uuid, sup in self.sup.items()}
+        return serve(endpoints, node_name or os.uname().nodename, port)
+
     def run(self, interval: float = 0.5):
         signal.signal(signal.SIGTERM, lambda *_: self.stop())
         while self.running:
@@ -192,6 +199,8 @@ def main():
     ap.add_argument("--window", type=float, default=C.WINDOW_MS)
     ap.add_argument("--inventory", default="auto",
                     choices=["auto", "amdsmi", "torch"])
+    ap.add_argument("--metrics-port", type=int, default=0,
+                    help="serve per-pod quota-enforcement metrics")
     args = ap.parse_args()
     daemon = NodeDaemon(args.workdir, args.base_port, args.base_quota,
                         args.min_quota, args.window, args.inventory)
@@ -199,6 +208,8 @@ def main():
         log("no GPUs found; exiting")
         return 1
     daemon.start()
+    if args.metrics_port:
+        daemon.serve_metrics(args.metrics_port)
     try:
         daemon.run()
     finally:

@@ -1,0 +1,61 @@
+"""Quota-enforcement observability: exports each local gpu-schd's STATS
+as Prometheus metrics (per-pod window usage, busy share, grants), so
+operators can see the actual enforced split next to the demand
+(gpu_requirement) and inventory (gpu_capacity) series. The reference
+only had Gemini's file logs here (SURVEY.md §5 tracing: none)."""
+from __future__ import annotations
+
+from prometheus_client.core import CounterMetricFamily, GaugeMetricFamily
+
+from ..isolation.client import query_stats
+
+
+class GpuSchdCollector:
+    """One collector over all of a node's gpu-schd daemons.
+
+    endpoints: {gpu_uuid: (host, port)}.
+    """
+
+    def __init__(self, endpoints: dict, node_name: str = ""):
+        self.endpoints = endpoints
+        self.node_name = node_name
+
+    def collect(self):
+        usage = GaugeMetricFamily(
+            "gpu_pod_window_usage_ms",
+            "pod GPU time inside the sliding window",
+            labels=["node", "uuid", "pod"])
+        share = GaugeMetricFamily(
+            "gpu_pod_busy_share",
+            "pod share of the GPU's busy time in the window",
+            labels=["node", "uuid", "pod"])
+        total = CounterMetricFamily(
+            "gpu_pod_total_used_ms",
+            "cumulative pod GPU time", labels=["node", "uuid", "pod"])
+        grants = CounterMetricFamily(
+            "gpu_pod_token_grants",
+            "cumulative token grants", labels=["node", "uuid", "pod"])
+        for uuid, (host, port) in self.endpoints.items():
+            try:
+                st = query_stats(host, port, timeout=2.0)
+            except OSError:
+                continue
+            for pod, v in st.get("pods", {}).items():
+                lab = [self.node_name, uuid, pod]
+                usage.add_metric(lab, v["usage_ms"])
+                share.add_metric(lab, v["busy_share"])
+                total.add_metric(lab, v["total_used_ms"])
+                grants.add_metric(lab, v["grants"])
+        yield usage
+        yield share
+        yield total
+        yield grants
+
+
+def serve(endpoints: dict, node_name: str, port: int):
+    from prometheus_client import CollectorRegistry, start_http_server
+
+    registry = CollectorRegistry()
+    registry.register(GpuSchdCollector(endpoints, node_name))
+    start_http_server(port, registry=registry)
+    return registry

@@ -229,3 +229,53 @@ def test_sharepod_conversion():
     assert pod["metadata"]["labels"][C.POD_GROUP_NAME] == "g"
     assert pod["spec"]["schedulerName"] == "kubeshare-scheduler"
     assert pod["metadata"]["ownerReferences"][0]["uid"] == "u-1"
+
+
+def test_noded_metrics_exporter(tmp_path, native_bins, fake_gpus):
+    """gpu-schd STATS surfaced as Prometheus metrics after real grants."""
+    base_port = 45000 + os.getpid() % 1000
+    nd = NodeDaemon(str(tmp_path), base_port=base_port, base_quota=50,
+                    min_quota=10, window=2000, gpus=fake_gpus[:1])
+    nd.start()
+    try:
+        ConfigDaemon("node-a", nd.config_dir, nd.port_dir).update(
+            [PodDemand("ns", "m1", "u1", "node-a", "GPU-fake-0",
+                       1.0, 0.5, 0, 50200)])
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            nd.poll_once()
+            try:
+                socket.create_connection(("127.0.0.1", 50200),
+                                         timeout=0.2).close()
+                break
+            except OSError:
+                time.sleep(0.1)
+        _token_roundtrip(50200, "x")
+        from kubeshare_amd.noded.metrics import GpuSchdCollector
+        reg = CollectorRegistry()
+        reg.register(GpuSchdCollector(
+            {"GPU-fake-0": ("127.0.0.1", base_port)}, "node-a"))
+        text = generate_latest(reg).decode()
+        assert 'gpu_pod_token_grants_total{node="node-a",pod="ns/m1",' \
+               'uuid="GPU-fake-0"}' in text
+        assert "gpu_pod_window_usage_ms" in text
+    finally:
+        nd.stop()
+
+
+def test_pod_group_gc():
+    import time as _t
+    from kubeshare_amd.scheduler.pod_group import PodGroupRegistry
+    reg = PodGroupRegistry(expiration_sec=0.1)
+    reg.get_or_create("ns", "g1", 100, 2)
+    assert "ns/g1" in reg.groups
+    reg.gc(now=_t.time() + 1.0)
+    assert "ns/g1" not in reg.groups
+
+
+def test_kube_modules_importable():
+    """The real-cluster drivers import without the kubernetes client
+    (it is only required at instantiation)."""
+    import kubeshare_amd.scheduler.kube  # noqa: F401
+    import kubeshare_amd.sharepod  # noqa: F401
+    import kubeshare_amd.queryip  # noqa: F401
