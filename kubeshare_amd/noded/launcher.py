@@ -128,7 +128,6 @@ class NodeDaemon:
         self.running = True
 
     def start(self):
-        q, m, w = self.knobs
         for gpu in self.gpus:
             uuid, idx = gpu["uuid"], gpu["index"]
             port = self.base_port + idx
@@ -138,16 +137,28 @@ class NodeDaemon:
             pf = os.path.join(self.port_dir, uuid)
             if not os.path.exists(pf):
                 F.write_port_config(self.port_dir, uuid, [])
-            self.schd[uuid] = subprocess.Popen(
-                [native_path("gpu-schd"), "-p", self.config_dir, "-f", uuid,
-                 "-P", str(port), "-q", str(q), "-m", str(m), "-w", str(w),
-                 "-l", os.path.join(self.log_dir, "gpu-schd.log")],
-                stderr=subprocess.DEVNULL)
+            self._spawn_schd(uuid, port)
             self.sup[uuid] = PodManagerSupervisor(uuid, port, self.log_dir)
-            log(f"gpu-schd started gpu={uuid} port={port}")
         return self
 
+    def _spawn_schd(self, uuid: str, port: int):
+        q, m, w = self.knobs
+        self.schd[uuid] = subprocess.Popen(
+            [native_path("gpu-schd"), "-p", self.config_dir, "-f", uuid,
+             "-P", str(port), "-q", str(q), "-m", str(m), "-w", str(w),
+             "-l", os.path.join(self.log_dir, "gpu-schd.log")],
+            stderr=subprocess.DEVNULL)
+        log(f"gpu-schd started gpu={uuid} port={port}")
+
     def poll_once(self):
+        # a crashed gpu-schd is restarted (SO_REUSEADDR: the port is
+        # rebindable immediately); hook clients reconnect on their next
+        # renewal and pod-mgrs are respawned below if they exited with it
+        for uuid, proc in list(self.schd.items()):
+            if proc.poll() is not None:
+                log(f"gpu-schd for {uuid} died (rc={proc.returncode}); "
+                    f"restarting")
+                self._spawn_schd(uuid, self.sup[uuid].sched_port)
         for uuid, sup in self.sup.items():
             path = os.path.join(self.port_dir, uuid)
             try:

@@ -279,3 +279,42 @@ def test_kube_modules_importable():
     import kubeshare_amd.scheduler.kube  # noqa: F401
     import kubeshare_amd.sharepod  # noqa: F401
     import kubeshare_amd.queryip  # noqa: F401
+
+
+def test_gpu_schd_crash_recovery(tmp_path, native_bins, fake_gpus):
+    """Kill a gpu-schd mid-run: the launcher restarts it, pod-mgr is
+    respawned, and a token round-trips again (restart-as-recovery, the
+    failure-detection property SURVEY.md §5 tracks)."""
+    base_port = 44000 + os.getpid() % 1000
+    nd = NodeDaemon(str(tmp_path), base_port=base_port, base_quota=50,
+                    min_quota=10, window=2000, gpus=fake_gpus[:1])
+    nd.start()
+    try:
+        ConfigDaemon("node-a", nd.config_dir, nd.port_dir).update(
+            [PodDemand("ns", "c1", "u1", "node-a", "GPU-fake-0",
+                       1.0, 0.5, 0, 50300)])
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            nd.poll_once()
+            try:
+                socket.create_connection(("127.0.0.1", 50300),
+                                         timeout=0.2).close()
+                break
+            except OSError:
+                time.sleep(0.1)
+        assert _token_roundtrip(50300, "x") > 0
+        # kill the scheduler daemon by its exact PID
+        nd.schd["GPU-fake-0"].kill()
+        nd.schd["GPU-fake-0"].wait()
+        deadline = time.time() + 10
+        ok = False
+        while time.time() < deadline and not ok:
+            nd.poll_once()
+            try:
+                if _token_roundtrip(50300, "x", timeout=3.0) > 0:
+                    ok = True
+            except OSError:
+                time.sleep(0.2)
+        assert ok, "token path did not recover after gpu-schd crash"
+    finally:
+        nd.stop()
