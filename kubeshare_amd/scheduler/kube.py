@@ -143,12 +143,13 @@ class KubeDriver:
         (reference pod.go:402-476)."""
         body = self.v1.read_namespaced_pod(pod.metadata.name,
                                            pod.metadata.namespace)
-        body.metadata.resource_version = None
+        original_uid = body.metadata.uid  # before nulling: `body` may be
+        body.metadata.resource_version = None  # the same cached object
         body.metadata.uid = None
         body.metadata.annotations = dict(body.metadata.annotations or {})
         body.metadata.annotations.update(placement.annotations)
         body.metadata.annotations["kubeshare.amd/original-uid"] = \
-            pod.metadata.uid
+            original_uid
         body.spec.node_name = placement.node
         env = [self.client.V1EnvVar(name=k, value=v)
                for k, v in placement.env.items()]
