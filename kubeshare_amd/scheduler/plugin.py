@@ -311,6 +311,7 @@ class KubeShareScheduler:
             C.POD_CELL_ID: leaf.id,
             C.POD_GPU_MEMORY: str(gpu_mem),
             C.POD_MANAGER_PORT: str(port),
+            C.POD_GPU_INDEX: str(idx),
         }
         env = {
             C.ENV_ROCR_VISIBLE_DEVICES: str(idx),
@@ -339,6 +340,7 @@ class KubeShareScheduler:
         ann = {
             C.POD_GPU_UUID: ",".join(uuids),
             C.POD_CELL_ID: ",".join(cells),
+            C.POD_GPU_INDEX: ",".join(map(str, idxs)),
         }
         env = {
             C.ENV_ROCR_VISIBLE_DEVICES: ",".join(map(str, idxs)),

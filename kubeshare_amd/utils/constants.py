@@ -29,6 +29,9 @@ POD_GPU_MODEL = DOMAIN + "gpu_model"
 POD_GPU_UUID = DOMAIN + "gpu_uuid"
 POD_CELL_ID = DOMAIN + "cell_id"
 POD_MANAGER_PORT = DOMAIN + "gpu_manager_port"
+# node-local device index(es) for ROCR_VISIBLE_DEVICES (consumed by the
+# mutating webhook, kubeshare_amd/webhook.py)
+POD_GPU_INDEX = "kubeshare.amd/gpu_index"
 
 SCHEDULER_NAME = "kubeshare-scheduler"
 
