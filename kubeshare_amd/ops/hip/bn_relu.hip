@@ -62,11 +62,14 @@ __global__ void bn_stats_kernel(const ushort8* __restrict__ x,
   const int cg = tid % CG;
   const int roff = tid / CG;
   const int rows_per_blk = KS_BN_BLOCK / CG;
+  // when CG does not divide the block, the tail threads would alias the
+  // NEXT block's rows — they must idle (their LDS slots stay zero)
+  const bool active = roff < rows_per_blk;
   const long long stride = (long long)gridDim.x * rows_per_blk;
 
   float8 acc = {0, 0, 0, 0, 0, 0, 0, 0};
   float8 acc2 = {0, 0, 0, 0, 0, 0, 0, 0};
-  long long row = (long long)blockIdx.x * rows_per_blk + roff;
+  long long row = active ? (long long)blockIdx.x * rows_per_blk + roff : M;
   // unrolled main loop: KS_BN_UNROLL_STATS independent loads in flight
   for (; row + (KS_BN_UNROLL_STATS - 1) * stride < M;
        row += KS_BN_UNROLL_STATS * stride) {
@@ -195,6 +198,7 @@ __global__ void bn_apply_relu_kernel(const ushort8* __restrict__ x,
   const int cg = tid % CG;
   const int roff = tid / CG;
   const int rows_per_blk = KS_BN_BLOCK / CG;
+  const bool active = roff < rows_per_blk;  // see bn_stats_kernel
   const long long stride = (long long)gridDim.x * rows_per_blk;
   float8 s, b;
 #pragma unroll
@@ -202,7 +206,7 @@ __global__ void bn_apply_relu_kernel(const ushort8* __restrict__ x,
     s[j] = scale[cg * 8 + j];
     b[j] = biasf[cg * 8 + j];
   }
-  long long row = (long long)blockIdx.x * rows_per_blk + roff;
+  long long row = active ? (long long)blockIdx.x * rows_per_blk + roff : M;
   for (; row + (KS_BN_UNROLL - 1) * stride < M;
        row += KS_BN_UNROLL * stride) {
     ushort8 v[KS_BN_UNROLL], r[KS_BN_UNROLL], o[KS_BN_UNROLL];
@@ -257,6 +261,7 @@ __global__ void bn_bwd_stats_kernel(const ushort8* __restrict__ x,
   const int cg = tid % CG;
   const int roff = tid / CG;
   const int rows_per_blk = KS_BN_BLOCK / CG;
+  const bool active = roff < rows_per_blk;  // see bn_stats_kernel
   const long long stride = (long long)gridDim.x * rows_per_blk;
   float8 mu, is;
 #pragma unroll
@@ -266,7 +271,7 @@ __global__ void bn_bwd_stats_kernel(const ushort8* __restrict__ x,
   }
   float8 adb = {0, 0, 0, 0, 0, 0, 0, 0};
   float8 ads = {0, 0, 0, 0, 0, 0, 0, 0};
-  long long row = (long long)blockIdx.x * rows_per_blk + roff;
+  long long row = active ? (long long)blockIdx.x * rows_per_blk + roff : M;
   for (; row + (KS_BN_UNROLL - 1) * stride < M;
        row += KS_BN_UNROLL * stride) {
     ushort8 xv[KS_BN_UNROLL], yv[KS_BN_UNROLL], gv[KS_BN_UNROLL];
@@ -351,6 +356,7 @@ __global__ void bn_bwd_apply_kernel(const ushort8* __restrict__ x,
   const int cg = tid % CG;
   const int roff = tid / CG;
   const int rows_per_blk = KS_BN_BLOCK / CG;
+  const bool active = roff < rows_per_blk;  // see bn_stats_kernel
   const long long stride = (long long)gridDim.x * rows_per_blk;
   const float invM = 1.f / (float)M;
   float8 mu, is, w, db, ds;
@@ -363,7 +369,7 @@ __global__ void bn_bwd_apply_kernel(const ushort8* __restrict__ x,
     db[j] = dbias[c] * invM;
     ds[j] = dscale[c] * invM;
   }
-  long long row = (long long)blockIdx.x * rows_per_blk + roff;
+  long long row = active ? (long long)blockIdx.x * rows_per_blk + roff : M;
   for (; row + (KS_BN_UNROLL - 1) * stride < M;
        row += KS_BN_UNROLL * stride) {
     ushort8 xv[KS_BN_UNROLL], yv[KS_BN_UNROLL], gv[KS_BN_UNROLL];

@@ -17,7 +17,10 @@ sys.path.insert(0, REPO)
 from kubeshare_amd import ops  # noqa: E402
 
 SHAPES = [(8, 64, 56, 56), (4, 256, 28, 28), (2, 2048, 7, 7),
-          (3, 8, 10, 10)]
+          (3, 8, 10, 10),
+          # CG (=C/8) NOT dividing the 256-thread block: exercises the
+          # inactive-thread guard (was a double-count bug)
+          (4, 24, 17, 17), (2, 1536, 9, 9)]
 
 
 def _mk(shape, seed=0):
