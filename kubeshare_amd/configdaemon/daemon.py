@@ -36,7 +36,8 @@ class ConfigDaemon:
                 continue
             pod = f"{d.namespace}/{d.name}"
             by_uuid_cfg.setdefault(d.uuid, []).append(
-                F.PodQuota(pod, d.limit, d.request, d.memory))
+                F.PodQuota(pod, d.limit, d.request, d.memory,
+                           group=d.group_name))
             by_uuid_port.setdefault(d.uuid, []).append(
                 F.PodPort(pod, d.port))
         if not by_uuid_cfg:

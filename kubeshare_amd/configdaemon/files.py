@@ -29,9 +29,11 @@ class PodQuota:
     limit: float
     request: float
     memory: int       # bytes; 0 = unlimited/default
+    group: str = ""   # gang group: members are co-granted by gpu-schd
 
     def line(self) -> str:
-        return f"{self.pod} {_fmt(self.limit)} {_fmt(self.request)} {self.memory}\n"
+        base = f"{self.pod} {_fmt(self.limit)} {_fmt(self.request)} {self.memory}"
+        return base + (f" {self.group}\n" if self.group else "\n")
 
 
 @dataclass(frozen=True)
@@ -79,9 +81,11 @@ def read_gpu_config(path: str) -> list[PodQuota]:
     n = int(lines[0])
     out = []
     for line in lines[1:1 + n]:
-        pod, limit, request, memory = line.split()
+        parts = line.split()
+        pod, limit, request, memory = parts[:4]
+        group = parts[4] if len(parts) > 4 else ""
         out.append(PodQuota(pod=pod, limit=float(limit), request=float(request),
-                            memory=int(memory)))
+                            memory=int(memory), group=group))
     return out
 
 

@@ -201,6 +201,9 @@ struct PodQuota {
   double limit = 1.0;
   double request = 0.0;
   long long memory = 0;
+  // optional 5th field: gang group — members are co-granted so a
+  // collective in one rank never spins on a token-starved peer
+  std::string group;
 };
 
 inline bool parse_gpu_config(FILE* f, std::vector<PodQuota>& out) {
@@ -212,10 +215,12 @@ inline bool parse_gpu_config(FILE* f, std::vector<PodQuota>& out) {
     if (!fgets(line, sizeof(line), f)) return false;
     PodQuota q;
     char pod[2048];
-    if (sscanf(line, "%2047s %lf %lf %lld", pod, &q.limit, &q.request,
-               &q.memory) != 4)
-      return false;
+    char group[1024] = "";
+    int got = sscanf(line, "%2047s %lf %lf %lld %1023s", pod, &q.limit,
+                     &q.request, &q.memory, group);
+    if (got < 4) return false;
     q.pod = pod;
+    if (got >= 5) q.group = group;
     out.push_back(q);
   }
   return true;

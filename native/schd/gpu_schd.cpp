@@ -107,29 +107,27 @@ int main(int argc, char** argv) {
   double wake_in = -1.0;  // ms until a capped waiter may become eligible
 
   auto flush_grants = [&]() {
+    // loop until no further grant is possible: gang co-granting means
+    // several same-group members can be granted back-to-back
     for (;;) {
       Grant g;
       double retry = 0.0;
-      if (sched.schedule(now_ms(), &g, &retry)) {
-        auto it = cookie2fd.find(g.cookie);
-        if (it != cookie2fd.end()) {
-          char line[64];
-          snprintf(line, sizeof(line), "GRANT %.3f", g.quota_ms);
-          if (!send_line(it->second, line)) {
-            // client went away between REQ and GRANT: free the token
-            sched.force_release(g.pod, now_ms());
-            cookie2fd.erase(it);
-            continue;
-          }
-          cookie2fd.erase(it);
-        } else {
-          sched.force_release(g.pod, now_ms());
-          continue;
-        }
-        break;  // one token outstanding
+      if (!sched.schedule(now_ms(), &g, &retry)) {
+        wake_in = retry > 0.0 ? retry : -1.0;
+        break;
       }
-      wake_in = retry > 0.0 ? retry : -1.0;
-      break;
+      auto it = cookie2fd.find(g.cookie);
+      if (it == cookie2fd.end()) {
+        sched.force_release(g.pod, now_ms());
+        continue;
+      }
+      char line[64];
+      snprintf(line, sizeof(line), "GRANT %.3f", g.quota_ms);
+      if (!send_line(it->second, line)) {
+        // client went away between REQ and GRANT: free the token
+        sched.force_release(g.pod, now_ms());
+      }
+      cookie2fd.erase(it);
     }
   };
 

@@ -44,6 +44,7 @@ struct PodAccount {
   double request = 0.0;
   double limit = 1.0;
   long long memory = 0;       // byte cap (0 = default)
+  std::string group;          // gang group ("" = none)
   bool in_config = false;     // listed in the per-UUID config file?
   // (t_charge_end_ms, used_ms) entries inside the window
   std::deque<std::pair<double, double>> charges;
@@ -92,6 +93,7 @@ class TokenScheduler {
       a.request = q.request;
       a.limit = q.limit;
       a.memory = q.memory;
+      a.group = q.group;
       a.in_config = true;
     }
   }
@@ -115,34 +117,41 @@ class TokenScheduler {
     waiters_.push_back(Waiter{pod, cookie, hint_ms, now});
   }
 
-  // The holder returned its token. `used_ms` is the client-measured GPU
-  // time; it is clamped server-side to [0, 3x quota + slack] so a lying
-  // or buggy client cannot starve others by under- or over-reporting
-  // less than it actually blocked the GPU (wall enforcement below
-  // catches over-holding).
+  // A holder returned its token. `used_ms` is the client-measured GPU
+  // time; it is clamped server-side to [0, wall + slack] so a lying
+  // client cannot under-report what it actually occupied.
   void release(const std::string& pod, double used_ms, double now) {
-    if (!holder_.empty() && pod == holder_) {
-      double wall = now - grant_ms_;
+    auto it = holders_.find(pod);
+    if (it != holders_.end()) {
+      double wall = now - it->second.grant_ms;
       double charged = std::max(0.0, std::min(used_ms, wall + 50.0));
       // a GPU-bound holder's wall time is the true exclusive occupancy
-      charged = std::max(charged, std::min(wall, grant_quota_));
+      charged = std::max(charged, std::min(wall, it->second.quota));
       account(pod).charge(now, charged);
-      holder_.clear();
+      holders_.erase(it);
+      if (holders_.empty()) holder_group_.clear();
     } else {
       // late RET after a revocation: already charged, ignore amount
     }
   }
 
-  // Holder liveness: revoke a token whose holder exceeded its lease by
-  // revoke_factor (it died or hung; its connection may still be open).
+  // Holder liveness: revoke tokens whose holders exceeded their lease
+  // (died or hung; the connection may still be open).
   bool check_revoke(double now, double* revoked_at = nullptr) {
-    if (holder_.empty()) return false;
-    double deadline = grant_ms_ + grant_quota_ * 3.0 + 2000.0;
-    if (now < deadline) return false;
-    account(holder_).charge(now, now - grant_ms_);
-    if (revoked_at) *revoked_at = now;
-    holder_.clear();
-    return true;
+    bool any = false;
+    for (auto it = holders_.begin(); it != holders_.end();) {
+      double deadline = it->second.grant_ms + it->second.quota * 3.0 + 2000.0;
+      if (now >= deadline) {
+        account(it->first).charge(now, now - it->second.grant_ms);
+        it = holders_.erase(it);
+        any = true;
+      } else {
+        ++it;
+      }
+    }
+    if (holders_.empty()) holder_group_.clear();
+    if (any && revoked_at) *revoked_at = now;
+    return any;
   }
 
   void drop_pod_waiters(long long cookie_lo, long long cookie_hi) {
@@ -156,24 +165,39 @@ class TokenScheduler {
 
   // If a holder vanished without RET (connection closed), charge wall.
   void force_release(const std::string& pod, double now) {
-    if (holder_ == pod) {
-      account(pod).charge(now, now - grant_ms_);
-      holder_.clear();
+    auto it = holders_.find(pod);
+    if (it != holders_.end()) {
+      account(pod).charge(now, now - it->second.grant_ms);
+      holders_.erase(it);
+      if (holders_.empty()) holder_group_.clear();
     }
   }
 
   // Pick the next holder. Returns true and fills `out` when a token was
   // granted; when false and `next_try_ms` > 0, every waiter is over its
-  // hard cap and the caller should re-run at now+next_try_ms.
+  // hard cap and the caller should re-run at now+next_try_ms. Called in
+  // a loop by the I/O layer, so GANG CO-GRANTING falls out naturally:
+  // while a gang's members hold the token, further members of the SAME
+  // group are granted immediately (a DDP collective in one rank must
+  // never spin on a token-starved peer — docs/ROADMAP.md), and everyone
+  // else waits until the whole gang has drained.
   bool schedule(double now, Grant* out, double* next_try_ms) {
     *next_try_ms = 0.0;
-    if (!holder_.empty() || waiters_.empty()) return false;
+    if (waiters_.empty()) return false;
+    const bool gang_active = !holders_.empty() && !holder_group_.empty();
+    if (!holders_.empty() && !gang_active) return false;
 
     double best_key = -1e300;
     int best_i = -1;
     double soonest = 1e300;
     for (int i = 0; i < (int)waiters_.size(); i++) {
       PodAccount& a = account(waiters_[i].pod);
+      if (!holders_.empty()) {
+        // only co-grant same-gang members (never the same pod twice)
+        if (a.group.empty() || a.group != holder_group_ ||
+            holders_.count(waiters_[i].pod))
+          continue;
+      }
       a.prune(now, window_);
       // limit >= 1.0 means "may use the whole GPU": no hard cap. (A
       // busy solo pod's window usage approaches window_ by definition;
@@ -211,9 +235,8 @@ class TokenScheduler {
                                    : a.limit * window_ - a.usage_cache;
     double quota = std::min(base_q_, room);
     quota = std::max(quota, min_q_);
-    holder_ = w.pod;
-    grant_ms_ = now;
-    grant_quota_ = quota;
+    holders_[w.pod] = Holder{now, quota};
+    holder_group_ = a.group;
     a.grants++;
     out->pod = w.pod;
     out->cookie = w.cookie;
@@ -221,7 +244,7 @@ class TokenScheduler {
     return true;
   }
 
-  const std::string& holder() const { return holder_; }
+  size_t n_holders() const { return holders_.size(); }
   size_t n_waiters() const { return waiters_.size(); }
 
   // One-line JSON stats: per-pod window usage, share of busy time,
@@ -261,12 +284,16 @@ class TokenScheduler {
     return b;
   }
 
+  struct Holder {
+    double grant_ms = 0.0;
+    double quota = 0.0;
+  };
+
   double base_q_, min_q_, window_;
   std::map<std::string, PodAccount> pods_;
   std::vector<Waiter> waiters_;
-  std::string holder_;
-  double grant_ms_ = 0.0;
-  double grant_quota_ = 0.0;
+  std::map<std::string, Holder> holders_;
+  std::string holder_group_;  // group of the current holders ("": none)
 };
 
 }  // namespace ks

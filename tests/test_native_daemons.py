@@ -286,3 +286,88 @@ def test_idle_release_work_conserving(native_bins, tmp_path):
         assert used_a > 250, (out_a, out_b)
     finally:
         schd.stop()
+
+
+def _write_config_grouped(tmp_path, pods):
+    """pods: (pod, limit, request, mem, group)."""
+    cfg = tmp_path / "config"
+    cfg.mkdir(exist_ok=True)
+    lines = [f"{len(pods)}"] + [
+        f"{pod} {limit} {request} {mem} {group}".rstrip()
+        for pod, limit, request, mem, group in pods
+    ]
+    (cfg / "GPU-x").write_text("\n".join(lines) + "\n")
+    return str(cfg)
+
+
+def test_gang_members_co_granted(native_bins, tmp_path):
+    """Two pods of one gang group are granted CONCURRENTLY (a DDP
+    collective in one rank must never spin on a token-starved peer),
+    while a third ungrouped pod waits for the whole gang to drain."""
+    cfg = _write_config_grouped(tmp_path, [
+        ("ns/g0", 1.0, 0.4, 0, "ddp"),
+        ("ns/g1", 1.0, 0.4, 0, "ddp"),
+        ("ns/solo", 1.0, 0.2, 0, ""),
+    ])
+    schd = Schd(native_bins, cfg, q=200, m=10, w=4000)
+    try:
+        from kubeshare_amd.isolation.client import TokenClient
+        a = TokenClient("127.0.0.1", schd.port, "ns/g0")
+        b = TokenClient("127.0.0.1", schd.port, "ns/g1")
+        qa = a.acquire()
+        assert qa > 0
+        # second gang member must be granted IMMEDIATELY while the
+        # first still holds (co-grant), not after qa expires
+        t0 = time.time()
+        qb = b.acquire()
+        assert qb > 0
+        assert time.time() - t0 < 1.0, "gang member was serialized"
+        st = schd.stats()
+        # both counted as holders: usage still zero, both granted once
+        assert st["pods"]["ns/g0"]["grants"] == 1
+        assert st["pods"]["ns/g1"]["grants"] == 1
+        a.release(50.0)
+        b.release(50.0)
+        a.close()
+        b.close()
+    finally:
+        schd.stop()
+
+
+def test_non_gang_waits_for_gang_drain(native_bins, tmp_path):
+    cfg = _write_config_grouped(tmp_path, [
+        ("ns/g0", 1.0, 0.4, 0, "ddp"),
+        ("ns/g1", 1.0, 0.4, 0, "ddp"),
+        ("ns/solo", 1.0, 0.2, 0, ""),
+    ])
+    schd = Schd(native_bins, cfg, q=200, m=10, w=4000)
+    try:
+        from kubeshare_amd.isolation.client import TokenClient
+        a = TokenClient("127.0.0.1", schd.port, "ns/g0")
+        b = TokenClient("127.0.0.1", schd.port, "ns/g1")
+        s = TokenClient("127.0.0.1", schd.port, "ns/solo")
+        a.acquire()
+        b.acquire()
+
+        import threading
+        got = {}
+
+        def solo_acquire():
+            got["t0"] = time.time()
+            got["quota"] = s.acquire()
+            got["t1"] = time.time()
+
+        th = threading.Thread(target=solo_acquire)
+        th.start()
+        time.sleep(0.4)
+        assert "quota" not in got, "solo pod granted while gang held"
+        a.release(100.0)
+        time.sleep(0.3)
+        assert "quota" not in got, "granted before the WHOLE gang drained"
+        b.release(100.0)
+        th.join(timeout=10)
+        assert got.get("quota", 0) > 0
+        for c in (a, b, s):
+            c.close()
+    finally:
+        schd.stop()
