@@ -175,10 +175,11 @@ class KubeDriver:
         print(f"[kubeshare-scheduler] {pod.metadata.namespace}/"
               f"{pod.metadata.name}: {reason}: {message}", flush=True)
 
-    def run(self, interval: float = 1.0):
+    def run(self, interval: float = 1.0, topology_path: str = ""):
         self.sync_nodes()
         self.resync_bound()
         last_gc = time.time()
+        topo_mtime = self._mtime(topology_path)
         while True:
             try:
                 self.run_once()
@@ -188,7 +189,23 @@ class KubeDriver:
                 self.sched.groups.gc()
                 self.sync_nodes()
                 last_gc = time.time()
+                # restart-as-reload on topology change (the reference
+                # exits and lets the Deployment restart it,
+                # config.go:122-136; in-memory state rebuilds from the
+                # bound-pod resync)
+                if topology_path and self._mtime(topology_path) != topo_mtime:
+                    print("[kubeshare-scheduler] topology changed; "
+                          "exiting for restart-as-reload", flush=True)
+                    raise SystemExit(0)
             time.sleep(interval)
+
+    @staticmethod
+    def _mtime(path: str):
+        import os
+        try:
+            return os.stat(path).st_mtime if path else None
+        except OSError:
+            return None
 
 
 def main():
@@ -203,7 +220,7 @@ def main():
     if args.fake_nodes:
         inv = FakeInventory({f"node-{i}": {"gpus": 8}
                              for i in range(args.fake_nodes)})
-    KubeDriver(topo, inv).run()
+    KubeDriver(topo, inv).run(topology_path=args.topology)
 
 
 if __name__ == "__main__":

@@ -59,6 +59,10 @@ def test_priority_domain():  # pod.go:179-199
         mk({C.POD_GPU_LIMIT: "1.0", C.POD_PRIORITY: "101"})
     with pytest.raises(LabelError):
         mk({C.POD_GPU_LIMIT: "1.0", C.POD_PRIORITY: "-2"})
+    # -1 is inside the reference's accepted range (pod.go:192) and is
+    # opportunistic
+    s = mk({C.POD_GPU_LIMIT: "1.0", C.POD_PRIORITY: "-1"})
+    assert s.is_opportunistic
 
 
 def test_garbage_values_rejected():
