@@ -43,6 +43,8 @@ class KubeDriver:
         self.watch = watch
         self.sched = KubeShareScheduler(topology)
         self.inventory = inventory
+        # gang members reserved but waiting at Permit: key -> (pod, placement)
+        self.waiting_pods: dict = {}
 
     # ------------------------------------------------------------ nodes
     def sync_nodes(self):
@@ -135,8 +137,6 @@ class KubeDriver:
                     self.apply_placement(w[0], w[1])
         else:
             self.waiting_pods[spec.key] = (pod, placement)
-
-    waiting_pods: dict = {}
 
     def apply_placement(self, pod, placement: Placement):
         """Shadow-pod recreate with injected env + pinned node
