@@ -371,3 +371,34 @@ def test_non_gang_waits_for_gang_drain(native_bins, tmp_path):
             c.close()
     finally:
         schd.stop()
+
+
+def test_unix_domain_socket_chain(native_bins, tmp_path):
+    """gpu-schd on a Unix socket + pod-mgr bridging UDS upstream to a
+    TCP client port — the no-hostNetwork deployment mode (SURVEY.md §5:
+    prefer UDS through the shared hostPath)."""
+    cfg = _write_config(tmp_path, [("ns/u1", 1.0, 0.5, 0)])
+    uds = str(tmp_path / "schd.sock")
+    schd = subprocess.Popen(
+        [native_bins["gpu-schd"], "-p", cfg, "-f", "GPU-x",
+         "-U", uds, "-q", "60", "-m", "10", "-w", "2000"],
+        stderr=subprocess.DEVNULL)
+    mgr_port = _free_port()
+    env = dict(os.environ, SCHEDULER_UDS=uds,
+               POD_MANAGER_PORT=str(mgr_port), POD_NAME="ns/u1")
+    mgr = None
+    try:
+        deadline = time.time() + 5
+        while time.time() < deadline and not os.path.exists(uds):
+            time.sleep(0.05)
+        mgr = subprocess.Popen([native_bins["pod-mgr"]], env=env,
+                               stderr=subprocess.DEVNULL)
+        _wait_listening(mgr_port)
+        res = _run_pods(native_bins, mgr_port, ["claimed"], 800)
+        assert res["claimed"][1] > 500
+    finally:
+        if mgr:
+            mgr.kill()
+            mgr.wait()
+        schd.kill()
+        schd.wait()

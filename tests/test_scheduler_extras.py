@@ -104,3 +104,17 @@ def test_resnet18_forward_cpu():
     from kubeshare_amd.models import build_model
     net = build_model("resnet18", num_classes=7)
     assert net(torch.randn(2, 3, 64, 64)).shape == (2, 7)
+
+
+def test_vgg16_fuse_model_sets_blocks():
+    from kubeshare_amd.models import build_model
+    from kubeshare_amd.models.vgg import ConvBNReLU
+    net = build_model("vgg16", num_classes=10)
+    blocks = [m for m in net.modules() if isinstance(m, ConvBNReLU)]
+    assert len(blocks) == 13  # VGG16's conv layers
+    assert not any(b.fused_ops for b in blocks)
+    # fuse_model flips the flag everywhere (ops ext import is required;
+    # on this CPU box the cross-compiled .so imports fine)
+    from kubeshare_amd import ops
+    ops.fuse_model(net)
+    assert all(b.fused_ops for b in blocks)
