@@ -427,6 +427,72 @@ hipError_t hipMemcpyWithStream(void* dst, const void* src, size_t n,
   return real(dst, src, n, k, stream);
 }
 
+hipError_t hipMemcpyHtoD(hipDeviceptr_t dst, const void* src, size_t n) {
+  REAL(hipMemcpyHtoD, hipDeviceptr_t, const void*, size_t);
+  if (!real) return hipErrorNotInitialized;
+  gate2(__builtin_return_address(0), nullptr);
+  return real(dst, src, n);
+}
+
+hipError_t hipMemcpyDtoH(void* dst, hipDeviceptr_t src, size_t n) {
+  REAL(hipMemcpyDtoH, void*, hipDeviceptr_t, size_t);
+  if (!real) return hipErrorNotInitialized;
+  gate2(__builtin_return_address(0), nullptr);
+  return real(dst, src, n);
+}
+
+hipError_t hipMemcpyDtoD(hipDeviceptr_t dst, hipDeviceptr_t src, size_t n) {
+  REAL(hipMemcpyDtoD, hipDeviceptr_t, hipDeviceptr_t, size_t);
+  if (!real) return hipErrorNotInitialized;
+  gate2(__builtin_return_address(0), nullptr);
+  return real(dst, src, n);
+}
+
+hipError_t hipMemcpyHtoDAsync(hipDeviceptr_t dst, const void* src, size_t n,
+                              hipStream_t stream) {
+  REAL(hipMemcpyHtoDAsync, hipDeviceptr_t, const void*, size_t, hipStream_t);
+  if (!real) return hipErrorNotInitialized;
+  gate2(__builtin_return_address(0), stream);
+  return real(dst, src, n, stream);
+}
+
+hipError_t hipMemcpyDtoHAsync(void* dst, hipDeviceptr_t src, size_t n,
+                              hipStream_t stream) {
+  REAL(hipMemcpyDtoHAsync, void*, hipDeviceptr_t, size_t, hipStream_t);
+  if (!real) return hipErrorNotInitialized;
+  gate2(__builtin_return_address(0), stream);
+  return real(dst, src, n, stream);
+}
+
+hipError_t hipMemcpyDtoDAsync(hipDeviceptr_t dst, hipDeviceptr_t src,
+                              size_t n, hipStream_t stream) {
+  REAL(hipMemcpyDtoDAsync, hipDeviceptr_t, hipDeviceptr_t, size_t,
+       hipStream_t);
+  if (!real) return hipErrorNotInitialized;
+  gate2(__builtin_return_address(0), stream);
+  return real(dst, src, n, stream);
+}
+
+hipError_t hipMemcpy2D(void* dst, size_t dpitch, const void* src,
+                       size_t spitch, size_t width, size_t height,
+                       hipMemcpyKind kind) {
+  REAL(hipMemcpy2D, void*, size_t, const void*, size_t, size_t, size_t,
+       hipMemcpyKind);
+  if (!real) return hipErrorNotInitialized;
+  gate2(__builtin_return_address(0), nullptr);
+  return real(dst, dpitch, src, spitch, width, height, kind);
+}
+
+hipError_t hipMemcpy2DAsync(void* dst, size_t dpitch, const void* src,
+                            size_t spitch, size_t width, size_t height,
+                            hipMemcpyKind kind, hipStream_t stream) {
+  REAL(hipMemcpy2DAsync, void*, size_t, const void*, size_t, size_t, size_t,
+       hipMemcpyKind, hipStream_t);
+  if (!real) return hipErrorNotInitialized;
+  gate2(__builtin_return_address(0), stream);
+  return real(dst, dpitch, src, spitch, width, height, kind, stream);
+}
+
 hipError_t hipMemset(void* dst, int value, size_t n) {
   REAL(hipMemset, void*, int, size_t);
   if (!real) return hipErrorNotInitialized;
