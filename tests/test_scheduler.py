@@ -293,3 +293,38 @@ def test_delete_pod_reclaims():
     leaf = fc.scheduler.tree.leaves_on_node("node-a")[0]
     assert leaf.available == 1.0
     assert leaf.free_memory == leaf.full_memory
+
+
+def test_restart_resync_multi_gpu_pod():
+    """Whole-GPU pods resync from their comma-joined uuid annotation
+    (reference pod.go:348-400 + processBoundPod)."""
+    fc = FakeCluster(nodes={"node-a": {"gpus": 4}})
+    pod = fc.add_pod("default", "mgr",
+                     {C.POD_GPU_REQUEST: "2.0", C.POD_GPU_LIMIT: "2.0"})
+    fc.schedule_pending()
+    assert pod.phase == "Bound"
+
+    fc2 = FakeCluster(nodes={"node-a": {"gpus": 4}})
+    err = fc2.scheduler.resync_bound_pod(
+        "default", "mgr", pod.labels, dict(pod.annotations), "node-a",
+        uid=pod.uid)
+    assert err is None
+    used = [c for c in fc2.scheduler.tree.leaves_on_node("node-a")
+            if c.available < 1.0]
+    assert len(used) == 2
+    assert all(c.available == 0.0 and c.free_memory == 0 for c in used)
+
+
+def test_periodic_inventory_resync_preserves_reservations():
+    """register_node is called periodically by the kube driver's
+    sync_nodes; re-registering the same inventory must NOT wipe live
+    reservations."""
+    fc = FakeCluster(nodes={"node-a": {"gpus": 2}})
+    pod = fc.add_pod("default", "keep", shared("0.5"))
+    fc.schedule_pending()
+    leaf = fc.scheduler.tree.leaf_by_uuid[pod.annotations[C.POD_GPU_UUID]]
+    assert leaf.available == pytest.approx(0.5)
+    fc.scheduler.register_node("node-a", fc.inventory.by_model("node-a"))
+    assert leaf.available == pytest.approx(0.5), "resync wiped reservation"
+    node_cell = fc.scheduler.tree.node_cells["node-a"][0]
+    assert node_cell.available == pytest.approx(1.5)
