@@ -83,7 +83,6 @@ def main():
     distributed = world > 1
 
     on_gpu = torch.cuda.is_available() and args.device != "cpu"
-    device = args.device or (f"cuda:{local_rank}" if on_gpu else "cpu")
 
     dist = None
     if distributed:
