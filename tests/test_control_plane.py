@@ -318,3 +318,39 @@ def test_gpu_schd_crash_recovery(tmp_path, native_bins, fake_gpus):
         assert ok, "token path did not recover after gpu-schd crash"
     finally:
         nd.stop()
+
+
+def test_launcher_pod_churn(tmp_path, native_bins, fake_gpus):
+    """Rapid add/remove of sharing pods: the launcher must converge to
+    the file state every time (no zombie pod-mgrs, no missed spawns)."""
+    base_port = 43000 + os.getpid() % 1000
+    nd = NodeDaemon(str(tmp_path), base_port=base_port, base_quota=40,
+                    min_quota=10, window=1500, gpus=fake_gpus[:1])
+    nd.start()
+    daemon = ConfigDaemon("node-a", nd.config_dir, nd.port_dir)
+    try:
+        for cycle in range(6):
+            pods = [PodDemand("ns", f"c{cycle}-{i}", f"u{i}", "node-a",
+                              "GPU-fake-0", 1.0, 0.3, 0, 50400 + i)
+                    for i in range(cycle % 3 + 1)]
+            daemon.update(pods)
+            deadline = time.time() + 10
+            want = {50400 + i for i in range(cycle % 3 + 1)}
+            while time.time() < deadline:
+                nd.poll_once()
+                if {nd.sup["GPU-fake-0"].ports[p]
+                        for p in nd.sup["GPU-fake-0"].procs} == want:
+                    break
+                time.sleep(0.05)
+            got = {nd.sup["GPU-fake-0"].ports[p]
+                   for p in nd.sup["GPU-fake-0"].procs}
+            assert got == want, f"cycle {cycle}: {got} != {want}"
+        daemon.update([])
+        for _ in range(40):
+            nd.poll_once()
+            if not nd.sup["GPU-fake-0"].procs:
+                break
+            time.sleep(0.05)
+        assert nd.sup["GPU-fake-0"].procs == {}
+    finally:
+        nd.stop()

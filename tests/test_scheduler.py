@@ -328,3 +328,38 @@ def test_periodic_inventory_resync_preserves_reservations():
     assert leaf.available == pytest.approx(0.5), "resync wiped reservation"
     node_cell = fc.scheduler.tree.node_cells["node-a"][0]
     assert node_cell.available == pytest.approx(1.5)
+
+
+def test_mixed_gpu_models_priority_and_pinning():
+    """A cluster mixing two GPU models: higher childCellPriority model
+    is preferred for unpinned pods (reference sortGPUPriority
+    cell.go:57-72), and gpu_model pins to the right leaves."""
+    from kubeshare_amd.scheduler.topology import CellSpec, CellTypeSpec
+    topo = TopologyConfig(
+        cell_types={
+            "MI355X-NODE": CellTypeSpec("AMD Instinct MI355X", 2, 200, True),
+            "MI300X-NODE": CellTypeSpec("AMD Instinct MI300X", 2, 100, True),
+        },
+        cells=[CellSpec(cell_type="MI355X-NODE", cell_id="fast-node"),
+               CellSpec(cell_type="MI300X-NODE", cell_id="slow-node")])
+    fc = FakeCluster(topology=topo,
+                     nodes={"fast-node": {"gpus": 2},
+                            "slow-node": {"gpus": 2,
+                                          "model": "AMD Instinct MI300X",
+                                          "memory": 192 * 1024**3}})
+    assert fc.scheduler.tree.models_by_priority[0] == "AMD Instinct MI355X"
+
+    # guarantee pod without a model pin lands on the faster model
+    p = fc.add_pod("default", "fastp",
+                   shared("0.5", **{C.POD_PRIORITY: "100"}))
+    fc.schedule_pending()
+    assert p.phase == "Bound" and p.node == "fast-node"
+
+    # pinned to the slower model
+    p2 = fc.add_pod("default", "slowp",
+                    shared("0.5", **{C.POD_GPU_MODEL:
+                                     "AMD Instinct MI300X"}))
+    fc.schedule_pending()
+    assert p2.phase == "Bound" and p2.node == "slow-node"
+    # default memory derives from THAT model's capacity
+    assert int(p2.annotations[C.POD_GPU_MEMORY]) == 96 * 1024**3
