@@ -436,3 +436,26 @@ def test_schd_survives_client_churn(native_bins, tmp_path):
         assert st["pods"]["ns/a"]["grants"] >= 1
     finally:
         schd.stop()
+
+
+def test_client_reconnects_after_schd_restart(native_bins, tmp_path):
+    """The hook-side TokenClient reconnects transparently when gpu-schd
+    restarts on the same port (SO_REUSEADDR + lazy reconnect)."""
+    cfg = _write_config(tmp_path, [("ns/r", 1.0, 0.5, 0)])
+    schd = Schd(native_bins, cfg, q=50, m=10, w=2000)
+    port = schd.port
+    try:
+        res = _run_pods(native_bins, port, ["ns/r"], 400)
+        assert res["ns/r"][1] > 200
+        schd.proc.kill()
+        schd.proc.wait()
+        # restart on the SAME port
+        schd.proc = subprocess.Popen(
+            [native_bins["gpu-schd"], "-p", cfg, "-f", "GPU-x",
+             "-P", str(port), "-q", "50", "-m", "10", "-w", "2000"],
+            stderr=subprocess.DEVNULL)
+        _wait_listening(port)
+        res = _run_pods(native_bins, port, ["ns/r"], 400)
+        assert res["ns/r"][1] > 200
+    finally:
+        schd.stop()
