@@ -118,3 +118,22 @@ def test_vgg16_fuse_model_sets_blocks():
     from kubeshare_amd import ops
     ops.fuse_model(net)
     assert all(b.fused_ops for b in blocks)
+
+
+def test_logger_format(tmp_path):
+    """Reference log format: "ts LEVEL: file:line msg" into
+    <dir>/<component>.log (pkg/logger/logger.go:40-57)."""
+    import re
+    from kubeshare_amd.utils.logger import get_logger
+    log = get_logger("testcomp", log_dir=str(tmp_path))
+    log.info("hello world")
+    text = (tmp_path / "testcomp.log").read_text()
+    assert re.search(
+        r"\d{4}-\d{2}-\d{2} \d{2}:\d{2}:\d{2}\.\d{3} INFO: "
+        r"test_scheduler_extras\.py:\d+ hello world", text), text
+
+
+def test_isolation_package_exports():
+    import kubeshare_amd.isolation as iso
+    assert hasattr(iso, "LocalGPUShare")
+    assert hasattr(iso, "TokenClient")
