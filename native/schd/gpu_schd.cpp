@@ -17,6 +17,7 @@
 
 #include <map>
 #include <memory>
+#include <set>
 
 #include "token_sched.hpp"
 
@@ -27,7 +28,7 @@ namespace {
 struct Client {
   int fd;
   LineBuffer rx;
-  long long cookie_lo, cookie_hi;  // cookie range issued to this conn
+  std::set<long long> pending;  // outstanding REQ cookies of this conn
   std::string last_pod;
 };
 
@@ -121,6 +122,8 @@ int main(int argc, char** argv) {
         sched.force_release(g.pod, now_ms());
         continue;
       }
+      auto cl = clients.find(it->second);
+      if (cl != clients.end()) cl->second->pending.erase(g.cookie);
       char line[64];
       snprintf(line, sizeof(line), "GRANT %.3f", g.quota_ms);
       if (!send_line(it->second, line)) {
@@ -151,8 +154,6 @@ int main(int argc, char** argv) {
         setsockopt(cfd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
         auto c = std::make_unique<Client>();
         c->fd = cfd;
-        c->cookie_lo = next_cookie;
-        c->cookie_hi = next_cookie - 1;  // empty range until first REQ
         clients[cfd] = std::move(c);
         if (clients.size() >= 1024) break;
       }
@@ -202,8 +203,7 @@ int main(int argc, char** argv) {
         if (tok[0] == "REQ" && tok.size() >= 2) {
           double hint = tok.size() >= 3 ? atof(tok[2].c_str()) : 0.0;
           long long ck = next_cookie++;
-          if (c.cookie_hi < c.cookie_lo) c.cookie_lo = ck;
-          c.cookie_hi = ck;
+          c.pending.insert(ck);
           c.last_pod = tok[1];
           cookie2fd[ck] = c.fd;
           sched.request(tok[1], ck, hint, now);
@@ -223,7 +223,7 @@ int main(int argc, char** argv) {
     for (int fd : dead) {
       Client& c = *clients[fd];
       // cancel outstanding REQs from this conn, release its token
-      sched.drop_pod_waiters(c.cookie_lo, c.cookie_hi);
+      sched.drop_waiters(c.pending);
       for (auto it = cookie2fd.begin(); it != cookie2fd.end();)
         it = (it->second == fd) ? cookie2fd.erase(it) : std::next(it);
       if (!c.last_pod.empty()) sched.force_release(c.last_pod, now_ms());

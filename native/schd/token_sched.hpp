@@ -33,6 +33,7 @@
 #include <algorithm>
 #include <deque>
 #include <map>
+#include <set>
 #include <string>
 #include <vector>
 
@@ -154,11 +155,13 @@ class TokenScheduler {
     return any;
   }
 
-  void drop_pod_waiters(long long cookie_lo, long long cookie_hi) {
+  // Drop the EXACT waiters a dead connection owns. (Cookies are global
+  // and interleave across connections — a range here once dropped other
+  // connections' pending REQs, stranding their clients.)
+  void drop_waiters(const std::set<long long>& cookies) {
     waiters_.erase(std::remove_if(waiters_.begin(), waiters_.end(),
                                   [&](const Waiter& w) {
-                                    return w.cookie >= cookie_lo &&
-                                           w.cookie <= cookie_hi;
+                                    return cookies.count(w.cookie) > 0;
                                   }),
                    waiters_.end());
   }
