@@ -267,9 +267,11 @@ class KubeShareScheduler:
         n_group = len(group_cells)
         scored = []
         need_whole = spec.is_multi_gpu
+        memory = spec.memory
         for c in leaves:
-            if need_whole and c.available < 1.0:
-                continue
+            if need_whole and (c.available < 1.0 or
+                               (memory > 0 and c.free_memory < memory)):
+                continue  # Filter counted only memory-satisfying leaves
             if spec.is_opportunistic:
                 s = c.priority + (0.0 if need_whole
                                   else (1.0 - c.available) * 100.0)

@@ -363,3 +363,23 @@ def test_mixed_gpu_models_priority_and_pinning():
     assert p2.phase == "Bound" and p2.node == "slow-node"
     # default memory derives from THAT model's capacity
     assert int(p2.annotations[C.POD_GPU_MEMORY]) == 96 * 1024**3
+
+
+def test_multi_gpu_memory_constraint_respected_at_reserve():
+    """A multi-GPU pod with a gpu_mem demand must only be placed on
+    leaves satisfying it — also at cell-selection time, not just in
+    Filter."""
+    fc = FakeCluster(nodes={"node-a": {"gpus": 4}})
+    # shrink two leaves' free memory below the demand
+    for c in fc.scheduler.tree.leaves_on_node("node-a")[:2]:
+        fc.scheduler.tree.reserve(c, 0.0, C.MI355X_HBM_BYTES // 2)
+    demand = str(C.MI355X_HBM_BYTES - 1024)
+    pod = fc.add_pod("default", "mgm",
+                     {C.POD_GPU_REQUEST: "2.0", C.POD_GPU_LIMIT: "2.0",
+                      C.POD_GPU_MEMORY: demand})
+    fc.schedule_pending()
+    assert pod.phase == "Bound"
+    picked = pod.annotations[C.POD_GPU_UUID].split(",")
+    full = {c.uuid for c in fc.scheduler.tree.leaves_on_node("node-a")
+            if c.free_memory >= int(demand) or c.uuid in picked}
+    assert set(picked) <= {"GPU-node-a-2", "GPU-node-a-3"}, picked
