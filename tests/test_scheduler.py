@@ -380,6 +380,19 @@ def test_multi_gpu_memory_constraint_respected_at_reserve():
     fc.schedule_pending()
     assert pod.phase == "Bound"
     picked = pod.annotations[C.POD_GPU_UUID].split(",")
-    full = {c.uuid for c in fc.scheduler.tree.leaves_on_node("node-a")
-            if c.free_memory >= int(demand) or c.uuid in picked}
     assert set(picked) <= {"GPU-node-a-2", "GPU-node-a-3"}, picked
+
+
+def test_guarantee_pods_win_scarce_capacity():
+    """QueueSort priority ordering: when capacity is scarce, Guarantee
+    pods are scheduled before Opportunistic ones in the same cycle
+    (reference Less, scheduler.go:247-267)."""
+    fc = FakeCluster(nodes={"node-a": {"gpus": 2}})
+    opps = [fc.add_pod("default", f"zo{i}", shared("1.0"))
+            for i in range(4)]  # names sort AFTER guarantee pods anyway
+    guas = [fc.add_pod("default", f"ag{i}",
+                       shared("1.0", **{C.POD_PRIORITY: "100"}))
+            for i in range(2)]
+    fc.schedule_pending()
+    assert all(p.phase == "Bound" for p in guas), [p.phase for p in guas]
+    assert all(p.phase != "Bound" for p in opps)
