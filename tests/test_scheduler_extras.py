@@ -137,3 +137,32 @@ def test_isolation_package_exports():
     import kubeshare_amd.isolation as iso
     assert hasattr(iso, "LocalGPUShare")
     assert hasattr(iso, "TokenClient")
+
+
+def test_fused_model_cpu_fallback_matches_stock():
+    """fuse_model'd ResNet on CPU routes through the eager fallback in
+    ops.bn_relu (no CUDA): outputs must match the stock module path."""
+    import torch
+    from kubeshare_amd import ops
+    from kubeshare_amd.models import resnet18
+    torch.manual_seed(0)
+    m1 = resnet18(num_classes=10)
+    m2 = resnet18(num_classes=10)
+    m2.load_state_dict(m1.state_dict())
+    ops.fuse_model(m2)
+    m1.eval()
+    m2.eval()
+    x = torch.randn(2, 3, 64, 64)
+    with torch.no_grad():
+        torch.testing.assert_close(m1(x), m2(x))
+
+
+def test_graft_entry_contract():
+    import importlib.util
+    import os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    spec = importlib.util.spec_from_file_location(
+        "graft_entry", os.path.join(repo, "__graft_entry__.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    assert callable(mod.build) and callable(mod.smoke)
