@@ -165,11 +165,19 @@ def test_full_stack_e2e(tmp_path, native_bins, fake_gpus):
                     min_quota=10, window=2000, gpus=gpus)
     nd.start()
     try:
-        # L3: aggregator demand -> config daemon -> files
+        # L3: aggregator demand -> config daemon -> files.
+        # The scheduler assigns pool ports (50050+); under parallel test
+        # runs another worker's ephemeral port can collide with that
+        # fixed range, so remap each demand to a freshly probed port
+        # (the pool-port plumbing itself is covered in test_scheduler).
+        from dataclasses import replace
+        from kubeshare_amd.isolation.local import free_port
         demands = [d for d in (demand_from_pod(p) for p in fc.pods.values())
                    if d is not None]
+        remap = {d.port: free_port() for d in demands}
+        demands = [replace(d, port=remap[d.port]) for d in demands]
         ConfigDaemon("node-a", nd.config_dir, nd.port_dir).update(demands)
-        ports = {p.key: int(p.annotations[C.POD_MANAGER_PORT])
+        ports = {p.key: remap[int(p.annotations[C.POD_MANAGER_PORT])]
                  for p in pods}
         deadline = time.time() + 10
         while time.time() < deadline:
