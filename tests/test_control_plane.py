@@ -107,26 +107,28 @@ def test_node_daemon_full_chain(tmp_path, native_bins, fake_gpus):
     """Launcher starts gpu-schd per GPU; config daemon publishes a pod;
     launcher spawns its pod-mgr; a token round-trips through the chain;
     removing the pod kills the pod-mgr."""
-    base_port = 47000 + os.getpid() % 1000
+    from kubeshare_amd.isolation.local import free_port
+    base_port = free_port()
     nd = NodeDaemon(str(tmp_path), base_port=base_port, base_quota=50,
                     min_quota=10, window=2000, gpus=fake_gpus)
     nd.start()
+    mgr_port = free_port()
     try:
         daemon = ConfigDaemon("node-a", nd.config_dir, nd.port_dir)
         daemon.update([PodDemand("ns", "p1", "u1", "node-a", "GPU-fake-0",
-                                 1.0, 0.5, 0, 50123)])
+                                 1.0, 0.5, 0, mgr_port)])
         deadline = time.time() + 10
         while time.time() < deadline:
             nd.poll_once()
             try:
-                socket.create_connection(("127.0.0.1", 50123),
+                socket.create_connection(("127.0.0.1", mgr_port),
                                          timeout=0.2).close()
                 break
             except OSError:
                 time.sleep(0.1)
         else:
             raise TimeoutError("pod-mgr never came up")
-        quota = _token_roundtrip(50123, "ignored")
+        quota = _token_roundtrip(mgr_port, "ignored")
         assert quota > 0
         # gpu-schd accounted it under the pod-mgr's stamped identity
         s = socket.create_connection(("127.0.0.1", base_port), timeout=5)
@@ -160,7 +162,8 @@ def test_full_stack_e2e(tmp_path, native_bins, fake_gpus):
     # L2 node daemon with inventory matching the scheduler's view
     gpus = [{"uuid": uuid, "model": C.MI355X_MODEL,
              "memory": C.MI355X_HBM_BYTES, "index": 0}]
-    base_port = 46000 + os.getpid() % 1000
+    from kubeshare_amd.isolation.local import free_port
+    base_port = free_port()
     nd = NodeDaemon(str(tmp_path), base_port=base_port, base_quota=40,
                     min_quota=10, window=2000, gpus=gpus)
     nd.start()
@@ -241,24 +244,26 @@ def test_sharepod_conversion():
 
 def test_noded_metrics_exporter(tmp_path, native_bins, fake_gpus):
     """gpu-schd STATS surfaced as Prometheus metrics after real grants."""
-    base_port = 45000 + os.getpid() % 1000
+    from kubeshare_amd.isolation.local import free_port
+    base_port = free_port()
     nd = NodeDaemon(str(tmp_path), base_port=base_port, base_quota=50,
                     min_quota=10, window=2000, gpus=fake_gpus[:1])
     nd.start()
+    mgr_port = free_port()
     try:
         ConfigDaemon("node-a", nd.config_dir, nd.port_dir).update(
             [PodDemand("ns", "m1", "u1", "node-a", "GPU-fake-0",
-                       1.0, 0.5, 0, 50200)])
+                       1.0, 0.5, 0, mgr_port)])
         deadline = time.time() + 10
         while time.time() < deadline:
             nd.poll_once()
             try:
-                socket.create_connection(("127.0.0.1", 50200),
+                socket.create_connection(("127.0.0.1", mgr_port),
                                          timeout=0.2).close()
                 break
             except OSError:
                 time.sleep(0.1)
-        _token_roundtrip(50200, "x")
+        _token_roundtrip(mgr_port, "x")
         from kubeshare_amd.noded.metrics import GpuSchdCollector
         reg = CollectorRegistry()
         reg.register(GpuSchdCollector(
@@ -293,24 +298,26 @@ def test_gpu_schd_crash_recovery(tmp_path, native_bins, fake_gpus):
     """Kill a gpu-schd mid-run: the launcher restarts it, pod-mgr is
     respawned, and a token round-trips again (restart-as-recovery, the
     failure-detection property SURVEY.md §5 tracks)."""
-    base_port = 44000 + os.getpid() % 1000
+    from kubeshare_amd.isolation.local import free_port
+    base_port = free_port()
     nd = NodeDaemon(str(tmp_path), base_port=base_port, base_quota=50,
                     min_quota=10, window=2000, gpus=fake_gpus[:1])
     nd.start()
+    mgr_port = free_port()
     try:
         ConfigDaemon("node-a", nd.config_dir, nd.port_dir).update(
             [PodDemand("ns", "c1", "u1", "node-a", "GPU-fake-0",
-                       1.0, 0.5, 0, 50300)])
+                       1.0, 0.5, 0, mgr_port)])
         deadline = time.time() + 10
         while time.time() < deadline:
             nd.poll_once()
             try:
-                socket.create_connection(("127.0.0.1", 50300),
+                socket.create_connection(("127.0.0.1", mgr_port),
                                          timeout=0.2).close()
                 break
             except OSError:
                 time.sleep(0.1)
-        assert _token_roundtrip(50300, "x") > 0
+        assert _token_roundtrip(mgr_port, "x") > 0
         # kill the scheduler daemon by its exact PID
         nd.schd["GPU-fake-0"].kill()
         nd.schd["GPU-fake-0"].wait()
@@ -319,7 +326,7 @@ def test_gpu_schd_crash_recovery(tmp_path, native_bins, fake_gpus):
         while time.time() < deadline and not ok:
             nd.poll_once()
             try:
-                if _token_roundtrip(50300, "x", timeout=3.0) > 0:
+                if _token_roundtrip(mgr_port, "x", timeout=3.0) > 0:
                     ok = True
             except OSError:
                 time.sleep(0.2)
@@ -331,19 +338,21 @@ def test_gpu_schd_crash_recovery(tmp_path, native_bins, fake_gpus):
 def test_launcher_pod_churn(tmp_path, native_bins, fake_gpus):
     """Rapid add/remove of sharing pods: the launcher must converge to
     the file state every time (no zombie pod-mgrs, no missed spawns)."""
-    base_port = 43000 + os.getpid() % 1000
+    from kubeshare_amd.isolation.local import free_port
+    base_port = free_port()
     nd = NodeDaemon(str(tmp_path), base_port=base_port, base_quota=40,
                     min_quota=10, window=1500, gpus=fake_gpus[:1])
     nd.start()
     daemon = ConfigDaemon("node-a", nd.config_dir, nd.port_dir)
+    churn_ports = [free_port() for _ in range(3)]
     try:
         for cycle in range(6):
             pods = [PodDemand("ns", f"c{cycle}-{i}", f"u{i}", "node-a",
-                              "GPU-fake-0", 1.0, 0.3, 0, 50400 + i)
+                              "GPU-fake-0", 1.0, 0.3, 0, churn_ports[i])
                     for i in range(cycle % 3 + 1)]
             daemon.update(pods)
             deadline = time.time() + 10
-            want = {50400 + i for i in range(cycle % 3 + 1)}
+            want = {churn_ports[i] for i in range(cycle % 3 + 1)}
             while time.time() < deadline:
                 nd.poll_once()
                 if {nd.sup["GPU-fake-0"].ports[p]
