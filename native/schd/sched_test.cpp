@@ -1,0 +1,197 @@
+// Deterministic unit tests for the TokenScheduler policy (virtual
+// clock — every TokenScheduler entry point takes `now`, so fairness,
+// caps, decay, revocation and gang co-granting are tested without a
+// single sleep). Run by tests/test_native_daemons.py::test_sched_unit.
+#include <cassert>
+#include <cstdio>
+
+#include "token_sched.hpp"
+
+using namespace ks;
+
+static int g_checks = 0;
+#define CHECK(cond)                                                       \
+  do {                                                                    \
+    if (!(cond)) {                                                        \
+      fprintf(stderr, "CHECK failed at %s:%d: %s\n", __FILE__, __LINE__,  \
+              #cond);                                                     \
+      return 1;                                                           \
+    }                                                                     \
+    g_checks++;                                                           \
+  } while (0)
+
+static PodQuota quota(const char* pod, double limit, double request,
+                      const char* group = "") {
+  PodQuota q;
+  q.pod = pod;
+  q.limit = limit;
+  q.request = request;
+  q.group = group;
+  return q;
+}
+
+int main() {
+  // ---- 1. solo pod: work-conserving regrant, usage accounting
+  {
+    TokenScheduler s(300, 20, 10000);
+    s.set_config({quota("a", 1.0, 0.5)});
+    Grant g;
+    double retry;
+    s.request("a", 1, 0, 0.0);
+    CHECK(s.schedule(0.0, &g, &retry));
+    CHECK(g.pod == "a" && g.quota_ms == 300);
+    CHECK(!s.schedule(0.0, &g, &retry));  // one holder, no co-grant (no gang)
+    s.release("a", 300, 300.0);
+    CHECK(s.n_holders() == 0);
+    s.request("a", 2, 0, 300.0);
+    CHECK(s.schedule(300.0, &g, &retry));  // immediate regrant
+    s.release("a", 300, 600.0);
+  }
+
+  // ---- 2. deficit fairness: the under-served pod wins
+  {
+    TokenScheduler s(100, 20, 10000);
+    s.set_config({quota("a", 1.0, 0.7), quota("b", 1.0, 0.3)});
+    double t = 0;
+    double used_a = 0, used_b = 0;
+    for (int i = 0; i < 100; i++) {
+      s.request("a", i * 2, 0, t);
+      s.request("b", i * 2 + 1, 0, t);
+      Grant g;
+      double retry;
+      CHECK(s.schedule(t, &g, &retry));
+      t += g.quota_ms;
+      (g.pod == "a" ? used_a : used_b) += g.quota_ms;
+      s.release(g.pod, g.quota_ms, t);
+      // drop the loser's stale waiter for the next round
+      s.drop_waiters({i * 2, i * 2 + 1});
+    }
+    double share_a = used_a / (used_a + used_b);
+    CHECK(share_a > 0.65 && share_a < 0.75);
+  }
+
+  // ---- 3. hard cap + window decay
+  {
+    TokenScheduler s(100, 20, 1000);  // 1 s window
+    s.set_config({quota("c", 0.4, 0.2)});
+    Grant g;
+    double retry;
+    double t = 0;
+    double granted = 0;
+    // burn up to the cap: 0.4 * 1000 = 400 ms of room
+    for (int i = 0; i < 10; i++) {
+      s.request("c", 100 + i, 0, t);
+      if (!s.schedule(t, &g, &retry)) break;
+      granted += g.quota_ms;
+      t += g.quota_ms;
+      s.release("c", g.quota_ms, t);
+    }
+    CHECK(granted >= 380 && granted <= 420);
+    // at the cap: refused with a retry hint
+    s.request("c", 200, 0, t);
+    CHECK(!s.schedule(t, &g, &retry));
+    CHECK(retry > 0);
+    // after the window decays, eligible again
+    CHECK(s.schedule(t + 1100.0, &g, &retry));
+    CHECK(g.pod == "c");
+  }
+
+  // ---- 4. min-quota clamp near the cap
+  {
+    TokenScheduler s(300, 20, 1000);
+    s.set_config({quota("d", 0.31, 0.31)});
+    Grant g;
+    double retry;
+    s.request("d", 1, 0, 0.0);
+    CHECK(s.schedule(0.0, &g, &retry));
+    CHECK(g.quota_ms == 300);  // room 310 -> full base quota
+    s.release("d", 300, 300.0);
+    s.request("d", 2, 0, 300.0);
+    CHECK(s.schedule(300.0, &g, &retry));
+    CHECK(g.quota_ms == 20);  // room 10 -> clamped up to min quota
+  }
+
+  // ---- 5. liveness revocation frees a wedged holder
+  {
+    TokenScheduler s(100, 20, 10000);
+    s.set_config({quota("a", 1.0, 0.5), quota("b", 1.0, 0.5)});
+    Grant g;
+    double retry;
+    s.request("a", 1, 0, 0.0);
+    CHECK(s.schedule(0.0, &g, &retry));
+    s.request("b", 2, 0, 10.0);
+    CHECK(!s.schedule(10.0, &g, &retry));
+    // deadline = quota*3 + 2000 = 2300
+    CHECK(!s.check_revoke(2200.0));
+    CHECK(s.check_revoke(2400.0));
+    CHECK(s.schedule(2400.0, &g, &retry));
+    CHECK(g.pod == "b");
+  }
+
+  // ---- 6. gang co-granting
+  {
+    TokenScheduler s(200, 20, 10000);
+    s.set_config({quota("g0", 1.0, 0.3, "ddp"), quota("g1", 1.0, 0.3, "ddp"),
+                  quota("solo", 1.0, 0.4)});
+    Grant g;
+    double retry;
+    s.request("g0", 1, 0, 0.0);
+    s.request("solo", 2, 0, 0.0);
+    s.request("g1", 3, 0, 1.0);
+    CHECK(s.schedule(1.0, &g, &retry));
+    std::string first = g.pod;
+    CHECK(first == "g0" || first == "solo");  // deficit order
+    if (first == "solo") {
+      // drain solo, then the gang pair must co-hold
+      s.release("solo", 200, 201.0);
+      CHECK(s.schedule(201.0, &g, &retry));
+      CHECK(g.pod == "g0" || g.pod == "g1");
+    }
+    // whoever of the gang holds, the partner is co-granted...
+    CHECK(s.schedule(202.0, &g, &retry));
+    CHECK(s.n_holders() == 2);
+    // ...and nobody else is until the WHOLE gang drains
+    s.request("solo", 4, 0, 203.0);
+    CHECK(!s.schedule(203.0, &g, &retry));
+    s.release("g0", 100, 300.0);
+    CHECK(!s.schedule(300.0, &g, &retry));  // g1 still holds
+    s.release("g1", 100, 310.0);
+    CHECK(s.schedule(310.0, &g, &retry));
+    CHECK(g.pod == "solo");
+  }
+
+  // ---- 7. exact-cookie waiter cancellation
+  {
+    TokenScheduler s(100, 20, 10000);
+    s.set_config({quota("a", 1.0, 0.5), quota("b", 1.0, 0.5)});
+    Grant g;
+    double retry;
+    s.request("a", 10, 0, 0.0);
+    CHECK(s.schedule(0.0, &g, &retry));   // a holds
+    s.request("a", 11, 0, 1.0);           // a's next REQ
+    s.request("b", 12, 0, 2.0);           // b waits
+    s.drop_waiters({11});                 // a's CONNECTION died
+    s.force_release("a", 3.0);
+    CHECK(s.schedule(3.0, &g, &retry));
+    CHECK(g.pod == "b" && g.cookie == 12);  // b unharmed
+  }
+
+  // ---- 8. uncapped limit >= 1.0 never degrades the quota
+  {
+    TokenScheduler s(300, 20, 1000);  // tiny window vs base quota
+    s.set_config({quota("e", 1.0, 0.5)});
+    Grant g;
+    double retry;
+    double t = 0;
+    for (int i = 0; i < 20; i++) {
+      s.request("e", i, 0, t);
+      CHECK(s.schedule(t, &g, &retry));
+      CHECK(g.quota_ms == 300);  // never clamped toward min quota
+      t += g.quota_ms;
+      s.release("e", g.quota_ms, t);
+    }
+  }
+
+  printf("sched_test OK (%d checks)\n", g_checks);
+  return 0;
+}

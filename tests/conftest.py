@@ -22,7 +22,8 @@ def _built(name: str) -> str:
 @pytest.fixture(scope="session")
 def native_bins():
     """Build (if needed) and return paths of the native daemons."""
-    targets = ["gpu-schd", "pod-mgr", "hook_selftest", "libhiphook.so"]
+    targets = ["gpu-schd", "pod-mgr", "hook_selftest", "sched_test",
+               "libhiphook.so"]
     if not all(os.path.exists(_built(t)) for t in targets):
         if shutil.which("make") is None:
             pytest.skip("make unavailable")

@@ -459,3 +459,13 @@ def test_client_reconnects_after_schd_restart(native_bins, tmp_path):
         assert res["ns/r"][1] > 200
     finally:
         schd.stop()
+
+
+def test_sched_unit(native_bins):
+    """Deterministic virtual-clock unit tests of the C++ token policy
+    (fairness, caps, decay, revocation, gang co-granting) — see
+    native/schd/sched_test.cpp."""
+    r = subprocess.run([native_bins["sched_test"]], capture_output=True,
+                       text=True, timeout=60)
+    assert r.returncode == 0, r.stderr or r.stdout
+    assert "sched_test OK" in r.stdout
