@@ -192,6 +192,69 @@ int main() {
     }
   }
 
+  // ---- 9. late RET after revocation is ignored (no double charge)
+  {
+    TokenScheduler s(100, 20, 10000);
+    s.set_config({quota("a", 1.0, 0.5)});
+    Grant g;
+    double retry;
+    s.request("a", 1, 0, 0.0);
+    CHECK(s.schedule(0.0, &g, &retry));
+    CHECK(s.check_revoke(5000.0));  // charged wall = 5000
+    double before = s.account("a").total_used_ms;
+    s.release("a", 400.0, 5100.0);  // late RET from the revoked holder
+    CHECK(s.account("a").total_used_ms == before);
+  }
+
+  // ---- 10. unknown pod runs opportunistically, then the config lands
+  {
+    TokenScheduler s(100, 20, 10000);
+    Grant g;
+    double retry;
+    s.request("ghost", 1, 0, 0.0);
+    CHECK(s.schedule(0.0, &g, &retry));  // request 0 / limit 1 defaults
+    CHECK(g.pod == "ghost");
+    s.release("ghost", 100, 100.0);
+    s.set_config({quota("ghost", 0.5, 0.5)});
+    CHECK(s.account("ghost").request == 0.5);
+    CHECK(s.account("ghost").in_config);
+    // earlier usage survives the reload
+    CHECK(s.account("ghost").total_used_ms == 100.0);
+  }
+
+  // ---- 11. a lying client cannot under-report its occupancy
+  {
+    TokenScheduler s(100, 20, 10000);
+    s.set_config({quota("liar", 1.0, 0.5)});
+    Grant g;
+    double retry;
+    s.request("liar", 1, 0, 0.0);
+    CHECK(s.schedule(0.0, &g, &retry));
+    // held 100 ms of wall but claims 1 ms: charged >= min(wall, quota)
+    s.release("liar", 1.0, 100.0);
+    CHECK(s.account("liar").total_used_ms >= 100.0);
+    // ...nor wildly over-report (clamped to wall + slack)
+    s.request("liar", 2, 0, 100.0);
+    CHECK(s.schedule(100.0, &g, &retry));
+    s.release("liar", 99999.0, 150.0);
+    CHECK(s.account("liar").total_used_ms <= 100.0 + 50.0 + 51.0);
+  }
+
+  // ---- 12. stats JSON carries the accounting fields
+  {
+    TokenScheduler s(100, 20, 10000);
+    s.set_config({quota("j", 0.8, 0.6)});
+    Grant g;
+    double retry;
+    s.request("j", 1, 0, 0.0);
+    CHECK(s.schedule(0.0, &g, &retry));
+    s.release("j", 100, 100.0);
+    std::string js = s.stats_json(100.0);
+    CHECK(js.find("\"j\"") != std::string::npos);
+    CHECK(js.find("\"request\":0.6000") != std::string::npos);
+    CHECK(js.find("\"grants\":1") != std::string::npos);
+  }
+
   printf("sched_test OK (%d checks)\n", g_checks);
   return 0;
 }
