@@ -150,7 +150,15 @@ class CellTree:
         if not leaf.is_leaf:
             raise ValueError("reserve/reclaim operate on leaf cells")
         whole_before = 1 if leaf.available >= 1.0 else 0
-        leaf.available = min(1.0, max(0.0, leaf.available + d_avail))
+        avail = min(1.0, max(0.0, leaf.available + d_avail))
+        # snap float drift: 1.0 - 0.3 - 0.1 + 0.1 + 0.3 ends at
+        # 0.9999999999999999, which would permanently hide the leaf
+        # from whole-GPU pods (property-tested in test_scheduler.py)
+        if abs(avail - 1.0) < 1e-9:
+            avail = 1.0
+        elif avail < 1e-9:
+            avail = 0.0
+        leaf.available = avail
         leaf.free_memory = max(0, min(leaf.full_memory,
                                       leaf.free_memory + d_mem))
         whole_after = 1 if leaf.available >= 1.0 else 0

@@ -396,3 +396,28 @@ def test_guarantee_pods_win_scarce_capacity():
     fc.schedule_pending()
     assert all(p.phase == "Bound" for p in guas), [p.phase for p in guas]
     assert all(p.phase != "Bound" for p in opps)
+
+
+def test_reserve_reclaim_float_drift():
+    """Fractional reserve/reclaim cycles must restore EXACT whole-GPU
+    availability (1.0 - 0.3 - 0.1 + 0.1 + 0.3 = 0.999... in doubles
+    would otherwise permanently leak whole-GPU capacity)."""
+    import itertools
+    fc = FakeCluster(nodes={"node-a": {"gpus": 1}})
+    tree = fc.scheduler.tree
+    leaf = tree.leaves_on_node("node-a")[0]
+    fracs = [0.1, 0.2, 0.25, 0.3, 0.33, 0.4, 0.5, 0.75]
+    for combo in itertools.combinations_with_replacement(fracs, 2):
+        if sum(combo) > 1.0:
+            continue
+        for r in combo:
+            tree.reserve(leaf, r, 0)
+        for r in combo:
+            tree.reclaim(leaf, r, 0)
+        assert leaf.available == 1.0, (combo, leaf.available)
+        assert leaf.available_whole == 1, combo
+    # a whole-GPU pod still fits after heavy fractional churn
+    pod = fc.add_pod("default", "whole",
+                     {C.POD_GPU_REQUEST: "1.0", C.POD_GPU_LIMIT: "1.0"})
+    fc.schedule_pending()
+    assert pod.phase == "Bound"
