@@ -150,11 +150,15 @@ def parse_pod(namespace: str, name: str, labels: dict, *, uid: str = "",
     request = 0.0
     if raw_request is not None:
         request = _parse_value(raw_request, C.POD_GPU_REQUEST)
-        if limit > 1.0 and limit != request:
-            raise LabelError(f"whole-GPU pods need limit == request "
-                             f"(limit={limit}, request={request})")
         if request > limit:
             raise LabelError(f"request {request} > limit {limit}")
+    # Conscious fix of a reference quirk (pod.go:249-298 only checks the
+    # pair when gpu_request is PRESENT): a whole-GPU limit with a
+    # missing/unequal request would otherwise produce a multi-GPU pod
+    # requesting 0 GPUs. Found by hypothesis (tests/test_properties.py).
+    if limit > 1.0 and limit != request:
+        raise LabelError(f"whole-GPU pods need limit == request "
+                         f"(limit={limit}, request={request})")
 
     if limit == 0.0 and request == 0.0:
         return None  # regular pod (pod.go:303-305)

@@ -1,0 +1,96 @@
+"""Property-based invariants (hypothesis): whatever random workload the
+scheduler sees, deleting every pod must restore the pristine cluster —
+no leaked availability, memory, or ports — and the label parser must
+never crash on arbitrary input."""
+import string
+
+from hypothesis import given, settings, strategies as st
+
+from kubeshare_amd.scheduler.harness import FakeCluster
+from kubeshare_amd.utils import constants as C
+from kubeshare_amd.utils.labels import LabelError, parse_pod
+
+FRACS = ["0.1", "0.2", "0.25", "0.3", "0.33", "0.5", "0.75", "1.0"]
+
+
+def pod_labels_strategy():
+    shared = st.builds(
+        lambda req, lim, prio: {
+            C.POD_GPU_REQUEST: req, C.POD_GPU_LIMIT: lim,
+            **({C.POD_PRIORITY: prio} if prio else {})},
+        st.sampled_from(FRACS), st.sampled_from(["1.0"]),
+        st.sampled_from(["", "0", "50", "100"]))
+    multi = st.builds(
+        lambda n: {C.POD_GPU_REQUEST: f"{n}.0", C.POD_GPU_LIMIT: f"{n}.0"},
+        st.integers(min_value=2, max_value=4))
+    limit_only = st.just({C.POD_GPU_LIMIT: "1.0"})
+    return st.one_of(shared, multi, limit_only)
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(pod_labels_strategy(), min_size=1, max_size=16),
+       st.randoms())
+def test_schedule_then_delete_restores_pristine_state(workload, rnd):
+    fc = FakeCluster(nodes={"node-a": {"gpus": 4}, "node-b": {"gpus": 4}})
+    tree = fc.scheduler.tree
+    pods = [fc.add_pod("p", f"pod{i}", labels)
+            for i, labels in enumerate(workload)]
+    fc.schedule_pending()
+    # delete in random order, interleaved with rescheduling attempts
+    order = list(pods)
+    rnd.shuffle(order)
+    for i, pod in enumerate(order):
+        fc.delete_pod(pod.key)
+        if i % 3 == 0:
+            fc.schedule_pending()
+    for node in ("node-a", "node-b"):
+        for leaf in tree.leaves_on_node(node):
+            assert leaf.available == 1.0, (leaf.id, leaf.available)
+            assert leaf.available_whole == 1
+            assert leaf.free_memory == leaf.full_memory, leaf.id
+        pool = fc.scheduler.ports[node]
+        assert all(not u for u in pool.used), "leaked manager port"
+    assert fc.scheduler.waiting == {}
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(pod_labels_strategy(), min_size=1, max_size=12))
+def test_never_oversubscribe_requests(workload):
+    """However pods are packed, the sum of reserved fractions per leaf
+    never exceeds 1.0 (requests are guarantees)."""
+    fc = FakeCluster(nodes={"node-a": {"gpus": 2}})
+    for i, labels in enumerate(workload):
+        fc.add_pod("q", f"pod{i}", labels)
+    fc.schedule_pending()
+    for leaf in fc.scheduler.tree.leaves_on_node("node-a"):
+        assert leaf.available >= 0.0, (leaf.id, leaf.available)
+        assert leaf.free_memory >= 0
+
+
+_label_text = st.text(
+    alphabet=string.ascii_letters + string.digits + "./-_", max_size=12)
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.dictionaries(
+    st.sampled_from([C.POD_GPU_REQUEST, C.POD_GPU_LIMIT, C.POD_GPU_MEMORY,
+                     C.POD_PRIORITY, C.POD_GPU_MODEL, C.POD_GROUP_NAME,
+                     C.POD_GROUP_HEADCOUNT, C.POD_GROUP_THRESHOLD,
+                     C.POD_MIN_AVAILABLE]),
+    _label_text, max_size=6))
+def test_label_parser_never_crashes(labels):
+    """Arbitrary label soup: parse_pod returns a spec, None, or raises
+    LabelError — never anything else; valid specs obey the invariants."""
+    try:
+        spec = parse_pod("ns", "p", labels)
+    except LabelError:
+        return
+    if spec is None:
+        return
+    assert spec.limit > 0
+    assert 0 <= spec.request <= spec.limit
+    if spec.limit > 1.0:
+        assert spec.limit == spec.request
+        assert float(spec.limit).is_integer()
+    assert spec.memory >= 0
+    assert -1 <= spec.priority <= 100
