@@ -94,3 +94,40 @@ def test_label_parser_never_crashes(labels):
         assert float(spec.limit).is_integer()
     assert spec.memory >= 0
     assert -1 <= spec.priority <= 100
+
+
+_k8s_name = st.from_regex(r"[a-z0-9]([a-z0-9.-]{0,10}[a-z0-9])?",
+                          fullmatch=True)
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.lists(
+    st.builds(lambda ns, n, lim, req, mem, grp: (f"{ns}/{n}", lim, req,
+                                                 mem, grp),
+              _k8s_name, _k8s_name,
+              st.floats(min_value=0.01, max_value=1.0),
+              st.floats(min_value=0.0, max_value=1.0),
+              st.integers(min_value=0, max_value=2**48),
+              st.one_of(st.just(""), _k8s_name)),
+    min_size=0, max_size=8))
+def test_config_file_roundtrip_property(tmp_path_factory, entries):
+    """Arbitrary (k8s-legal) pods/groups/values survive the per-UUID
+    file write/read round trip bit-exactly enough for scheduling
+    (request/limit to 1e-6, memory and group exactly)."""
+    from kubeshare_amd.configdaemon import files as F
+    tmp = tmp_path_factory.mktemp("cfgprop")
+    quotas = [F.PodQuota(pod, min(1.0, max(req, 0.0) if req <= lim else lim),
+                         lim, mem, group=grp)
+              for pod, lim, req, mem, grp in entries]
+    # kwargs order in PodQuota: (pod, limit, request, memory, group)
+    quotas = [F.PodQuota(pod, lim, min(req, lim), mem, group=grp)
+              for pod, lim, req, mem, grp in entries]
+    path = F.write_gpu_config(str(tmp), "GPU-prop", quotas)
+    back = F.read_gpu_config(path)
+    assert len(back) == len(quotas)
+    for a, b in zip(quotas, back):
+        assert a.pod == b.pod
+        assert abs(a.limit - b.limit) < 1e-6
+        assert abs(a.request - b.request) < 1e-6
+        assert a.memory == b.memory
+        assert a.group == b.group
