@@ -116,10 +116,7 @@ def test_config_file_roundtrip_property(tmp_path_factory, entries):
     (request/limit to 1e-6, memory and group exactly)."""
     from kubeshare_amd.configdaemon import files as F
     tmp = tmp_path_factory.mktemp("cfgprop")
-    quotas = [F.PodQuota(pod, min(1.0, max(req, 0.0) if req <= lim else lim),
-                         lim, mem, group=grp)
-              for pod, lim, req, mem, grp in entries]
-    # kwargs order in PodQuota: (pod, limit, request, memory, group)
+    # PodQuota(pod, limit, request, memory, group); request <= limit
     quotas = [F.PodQuota(pod, lim, min(req, lim), mem, group=grp)
               for pod, lim, req, mem, grp in entries]
     path = F.write_gpu_config(str(tmp), "GPU-prop", quotas)
