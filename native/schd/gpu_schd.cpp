@@ -71,10 +71,19 @@ int main(int argc, char** argv) {
 
   TokenScheduler sched(base_q, min_q, window);
 
+  // Order matters: install the config WATCH before anyone can observe
+  // us as ready (the listen socket), then load — a config written
+  // between an initial load and a later watch would be missed forever.
   std::string config_path;
+  int ino_fd = -1;
   if (!config_dir.empty() && !config_file.empty()) {
     if (config_dir.back() != '/') config_dir.push_back('/');
     config_path = config_dir + config_file;
+    ino_fd = inotify_init1(IN_NONBLOCK);
+    if (ino_fd >= 0) {
+      (void)inotify_add_watch(ino_fd, config_dir.c_str(),
+                              IN_CLOSE_WRITE | IN_MOVED_TO | IN_CREATE);
+    }
     std::vector<PodQuota> quotas;
     if (load_gpu_config(config_path.c_str(), quotas)) {
       sched.set_config(quotas);
@@ -91,16 +100,6 @@ int main(int argc, char** argv) {
   }
   logf(g_log, "gpu-schd", "listening (%s port %d) q=%.0f m=%.0f w=%.0f",
        uds_path.c_str(), port, base_q, min_q, window);
-
-  int ino_fd = -1, ino_wd = -1;
-  if (!config_dir.empty()) {
-    ino_fd = inotify_init1(IN_NONBLOCK);
-    if (ino_fd >= 0) {
-      ino_wd = inotify_add_watch(ino_fd, config_dir.c_str(),
-                                 IN_CLOSE_WRITE | IN_MOVED_TO | IN_CREATE);
-      (void)ino_wd;
-    }
-  }
 
   std::map<int, std::unique_ptr<Client>> clients;
   std::map<long long, int> cookie2fd;  // outstanding REQ cookie -> client fd
