@@ -38,3 +38,7 @@ class RRPortPool:
         idx = port - self.base
         if 0 <= idx < self.size:
             self.used[idx] = False
+
+    def is_free(self, port: int) -> bool:
+        idx = port - self.base
+        return 0 <= idx < self.size and not self.used[idx]

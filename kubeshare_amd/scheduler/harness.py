@@ -164,21 +164,16 @@ class FakeCluster:
         (reference Unreserve rejects all waiting members,
         scheduler.go:534-549)."""
         sch = self.scheduler
-        for key, waiters in list(sch.waiting.items()):
-            expired = [k for k, dl in waiters.items() if dl <= self.clock]
-            if not expired:
-                continue
-            for k in list(waiters.keys()):
-                pod = self.pods.get(k)
-                spec = sch.pod_status.get(k)
-                if spec is not None:
-                    sch.unreserve(spec)
-                if pod is not None:
-                    pod.phase = "Unschedulable"
-                    for a in (C.POD_GPU_UUID, C.POD_CELL_ID,
-                              C.POD_GPU_MEMORY, C.POD_MANAGER_PORT):
-                        pod.annotations.pop(a, None)
-            sch.waiting.pop(key, None)
+        for k in sch.expired_waiting(self.clock):
+            pod = self.pods.get(k)
+            spec = sch.pod_status.get(k)
+            if spec is not None:
+                sch.unreserve(spec)
+            if pod is not None:
+                pod.phase = "Unschedulable"
+                for a in (C.POD_GPU_UUID, C.POD_CELL_ID,
+                          C.POD_GPU_MEMORY, C.POD_MANAGER_PORT):
+                    pod.annotations.pop(a, None)
 
     def advance(self, seconds: float):
         self.clock += seconds

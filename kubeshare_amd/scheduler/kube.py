@@ -10,15 +10,22 @@ PreFilter/Filter/Score/Reserve/Permit, then APPLY the placement.
 Applying a placement needs env injection, which k8s forbids on an
 existing pod — the reference deletes and recreates a "shadow pod"
 (pkg/scheduler/scheduler.go:515-528, pod.go:402-476). We keep that
-mechanism for parity (recreate_with_injection) but preserve the
-original UID linkage via an annotation so controllers can correlate;
-a mutating-webhook flow (inject at create time from the
+mechanism for parity (apply_placement) but preserve the original UID
+linkage via an annotation so controllers can correlate; a
+mutating-webhook flow (inject at create time from the
 sharedgpu/gpu_uuid annotation) is the documented alternative
-(SURVEY.md §7 phase 1.3).
+(SURVEY.md §7 phase 1.3) and shares the same env builder
+(webhook.shared_pod_env) so both paths inject identically.
 
-Inventory comes from the collector's amdsmi export (or node labels),
-injected as a provider — never from Prometheus inside the Filter hot
-path (reference flaw, README.md:141).
+Client: the official `kubernetes` package when importable, else the
+stdlib REST client (kubeclient.py) — which also runs against the
+in-process fake API server in CI (tests/test_kube_e2e.py).
+
+Gang semantics: pods parked at Permit are tracked with a deadline
+(2 s x group headcount, reference scheduler.go:44,573); the run loop
+expires them, unreserving the whole gang (reference Unreserve rejects
+all waiting members, scheduler.go:534-549) — a partial gang never
+leaks its reservations.
 """
 from __future__ import annotations
 
@@ -26,21 +33,29 @@ import argparse
 import time
 
 from ..utils import constants as C
+from ..webhook import SHARED_VOLUMES, shared_pod_env
 from .inventory import FakeInventory
 from .plugin import KubeShareScheduler, Placement
 from .topology import TopologyConfig
 
 
-class KubeDriver:
-    def __init__(self, topology: TopologyConfig, inventory=None):
-        from kubernetes import client, config, watch  # lazy: optional dep
+def make_client():
+    """Official client when available, stdlib REST client otherwise."""
+    try:
+        from kubernetes import client, config
         try:
             config.load_incluster_config()
         except Exception:  # noqa: BLE001
             config.load_kube_config()
-        self.v1 = client.CoreV1Api()
-        self.client = client
-        self.watch = watch
+        return client.CoreV1Api()
+    except ImportError:
+        from .kubeclient import RestCoreV1
+        return RestCoreV1()
+
+
+class KubeDriver:
+    def __init__(self, topology: TopologyConfig, inventory=None, api=None):
+        self.v1 = api if api is not None else make_client()
         self.sched = KubeShareScheduler(topology)
         self.inventory = inventory
         # gang members reserved but waiting at Permit: key -> (pod, placement)
@@ -89,13 +104,44 @@ class KubeDriver:
 
     # ---------------------------------------------------------- the loop
     def run_once(self):
+        self.expire_permits()
         pods = [p for p in self.v1.list_pod_for_all_namespaces(
                     field_selector="status.phase=Pending").items
                 if p.spec.scheduler_name == C.SCHEDULER_NAME
                 and not p.spec.node_name]
-        for p in sorted(pods,
-                        key=lambda p: p.metadata.creation_timestamp or 0):
+        for p in sorted(pods, key=self._queue_key):
+            if f"{p.metadata.namespace}/{p.metadata.name}" in \
+                    self.waiting_pods:
+                continue  # parked at Permit; released or expired, not rerun
             self.schedule_pod(p)
+
+    def _queue_key(self, pod):
+        """Reference QueueSort (scheduler.go:247-267): group priority
+        desc, then creation time, then key."""
+        prio = 0
+        try:
+            from ..utils.labels import parse_pod
+            spec = parse_pod(pod.metadata.namespace, pod.metadata.name,
+                             pod.metadata.labels or {})
+            if spec is not None:
+                prio = spec.priority
+        except Exception:  # noqa: BLE001 — invalid labels sort last
+            prio = -1000
+        return (-prio, str(pod.metadata.creation_timestamp or ""),
+                f"{pod.metadata.namespace}/{pod.metadata.name}")
+
+    def expire_permits(self):
+        """Reject gangs whose Permit wait expired; their members were
+        never applied to the cluster, so only local reservations (cells,
+        port) must be reclaimed."""
+        for key in self.sched.expired_waiting():
+            spec = self.sched.pod_status.get(key)
+            entry = self.waiting_pods.pop(key, None)
+            if spec is not None:
+                self.sched.unreserve(spec)
+            if entry is not None:
+                self.event(entry[0], "GangTimeout",
+                           "Permit wait expired; gang rejected")
 
     def schedule_pod(self, pod):
         ns, name = pod.metadata.namespace, pod.metadata.name
@@ -138,6 +184,17 @@ class KubeDriver:
         else:
             self.waiting_pods[spec.key] = (pod, placement)
 
+    def placement_env(self, pod, placement: Placement) -> list:
+        """Env entries (plain dicts — both clients serialize them) for
+        the injected containers; identical to the webhook path."""
+        idx = ",".join(map(str, placement.gpu_indices))
+        if placement.manager_port:
+            return shared_pod_env(pod.metadata.namespace, pod.metadata.name,
+                                  str(placement.manager_port),
+                                  str(placement.gpu_mem), idx)
+        return [{"name": C.ENV_INJECTED, "value": "1"},
+                {"name": C.ENV_ROCR_VISIBLE_DEVICES, "value": idx}]
+
     def apply_placement(self, pod, placement: Placement):
         """Shadow-pod recreate with injected env + pinned node
         (reference pod.go:402-476)."""
@@ -146,26 +203,25 @@ class KubeDriver:
         original_uid = body.metadata.uid  # before nulling: `body` may be
         body.metadata.resource_version = None  # the same cached object
         body.metadata.uid = None
-        body.metadata.annotations = dict(body.metadata.annotations or {})
-        body.metadata.annotations.update(placement.annotations)
-        body.metadata.annotations["kubeshare.amd/original-uid"] = \
-            original_uid
+        ann = dict((body.metadata.annotations or {}).items())
+        ann.update(placement.annotations)
+        ann["kubeshare.amd/original-uid"] = original_uid
+        body.metadata.annotations = ann
         body.spec.node_name = placement.node
-        env = [self.client.V1EnvVar(name=k, value=v)
-               for k, v in placement.env.items()]
-        for container in body.spec.containers:
+        env = self.placement_env(pod, placement)
+        containers = body.spec.containers
+        for container in containers:
             container.env = (container.env or []) + env
             if placement.manager_port:
                 container.volume_mounts = (container.volume_mounts or []) + [
-                    self.client.V1VolumeMount(
-                        name="kubeshare-library",
-                        mount_path=C.LIBRARY_PATH)]
+                    {"name": "kubeshare-library",
+                     "mountPath": C.LIBRARY_PATH, "readOnly": True},
+                    {"name": "kubeshare-sock", "mountPath": C.SOCK_DIR},
+                ]
+        body.spec.containers = containers
         if placement.manager_port:
-            body.spec.volumes = (body.spec.volumes or []) + [
-                self.client.V1Volume(
-                    name="kubeshare-library",
-                    host_path=self.client.V1HostPathVolumeSource(
-                        path=C.LIBRARY_PATH))]
+            body.spec.volumes = (body.spec.volumes or []) + \
+                [dict(v) for v in SHARED_VOLUMES]
         self.v1.delete_namespaced_pod(pod.metadata.name,
                                       pod.metadata.namespace,
                                       grace_period_seconds=0)
@@ -214,13 +270,19 @@ def main():
                     default=C.CLUSTER_TOPOLOGY_FILE)
     ap.add_argument("--fake-nodes", type=int, default=0,
                     help="dev mode: N fake 8-GPU nodes instead of amdsmi")
+    ap.add_argument("--api-url", default="",
+                    help="API server base URL (stdlib REST client)")
     args = ap.parse_args()
     topo = TopologyConfig.from_file(args.topology)
     inv = None
     if args.fake_nodes:
         inv = FakeInventory({f"node-{i}": {"gpus": 8}
                              for i in range(args.fake_nodes)})
-    KubeDriver(topo, inv).run(topology_path=args.topology)
+    api = None
+    if args.api_url:
+        from .kubeclient import RestCoreV1
+        api = RestCoreV1(args.api_url)
+    KubeDriver(topo, inv, api=api).run(topology_path=args.topology)
 
 
 if __name__ == "__main__":

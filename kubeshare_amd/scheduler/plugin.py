@@ -406,13 +406,31 @@ class KubeShareScheduler:
             release = list(waiting.keys())
             self.waiting.pop(key, None)
             return "allow", 0.0, release
-        timeout = self.permit_waiting_time * max(spec.min_available, 1)
+        # timeout unit is the group HEADCOUNT (reference scheduler.go:44,
+        # 573: 2 s x headcount); min_available is only the fallback when
+        # the gang was declared via the direct min_available label
+        gang_size = spec.headcount if spec.headcount > 0 else spec.min_available
+        timeout = self.permit_waiting_time * max(gang_size, 1)
         waiting[spec.key] = (now if now is not None else time.time()) + timeout
         return "wait", timeout, []
 
     def reject_waiting_group(self, namespace: str, group: str) -> list:
         key = self.groups.key_for(namespace, group)
         return list(self.waiting.pop(key, {}).keys())
+
+    def expired_waiting(self, now: float | None = None) -> list:
+        """Pod keys of gangs whose Permit wait has expired. When ANY
+        member's deadline passes, the WHOLE waiting gang is rejected
+        (reference Unreserve rejects all waiting group members,
+        scheduler.go:534-549). The caller must unreserve each returned
+        pod; the waiting entries are popped here."""
+        t = now if now is not None else time.time()
+        out = []
+        for key, waiters in list(self.waiting.items()):
+            if any(dl <= t for dl in waiters.values()):
+                out.extend(waiters.keys())
+                self.waiting.pop(key, None)
+        return out
 
     # -------------------------------------------------------- restart sync
     def resync_bound_pod(self, namespace: str, name: str, labels: dict,

@@ -43,6 +43,10 @@ ENV_HIP_VISIBLE_DEVICES = "HIP_VISIBLE_DEVICES"
 ENV_LD_PRELOAD = "LD_PRELOAD"
 ENV_POD_MANAGER_IP = "POD_MANAGER_IP"
 ENV_POD_MANAGER_PORT = "POD_MANAGER_PORT"
+ENV_POD_MANAGER_UDS = "POD_MANAGER_UDS"
+# marker the webhook keys idempotency on (a user-set ROCR_VISIBLE_DEVICES
+# must NOT suppress injection — that would silently skip the hook)
+ENV_INJECTED = "KUBESHARE_INJECTED"
 ENV_POD_NAME = "POD_NAME"
 ENV_SCHEDULER_IP = "SCHEDULER_IP"
 ENV_SCHEDULER_PORT = "SCHEDULER_PORT"
@@ -57,6 +61,18 @@ LIBRARY_PATH = KUBESHARE_ROOT + "/library"
 HOOK_SO_NAME = "libhiphook.so"
 HOOK_SO_PATH = LIBRARY_PATH + "/" + HOOK_SO_NAME
 SCHEDULER_IP_FILE = LIBRARY_PATH + "/schedulerIP.txt"
+# per-pod pod-mgr unix sockets live here (hostPath mounted RW into the
+# container; a UDS connect() needs write access to the socket inode, so
+# this cannot share the read-only /kubeshare/library mount). Default
+# transport hook<->pod-mgr: UDS keyed by the pod's manager port — no
+# hostNetwork, unreachable from other pods that don't mount it.
+SOCK_DIR = KUBESHARE_ROOT + "/sock"
+
+
+def pod_manager_uds(port: int, root: str = SOCK_DIR) -> str:
+    """Socket path for one pod's manager, keyed by its allocated port
+    (ports are already unique per node via the 512-wide pool)."""
+    return f"{root}/pm-{port}.sock"
 LOG_PATH = KUBESHARE_ROOT + "/log"
 SCHEDULER_CONFIG_ROOT = KUBESHARE_ROOT + "/scheduler"
 GPU_CONFIG_DIR = SCHEDULER_CONFIG_ROOT + "/config/"

@@ -45,6 +45,7 @@ class PodSpec:
     model: str = ""
     pod_group: str = ""
     min_available: int = 0
+    headcount: int = 0  # sharedgpu/group_headcount (Permit timeout unit)
     # filled at Reserve time
     uuids: list = field(default_factory=list)
     cell_ids: list = field(default_factory=list)
@@ -93,33 +94,43 @@ def parse_priority(labels: dict) -> int:
     return p
 
 
-def parse_gang(labels: dict) -> tuple[str, int]:
-    """Returns (group_name, min_available).
+def parse_gang(labels: dict) -> tuple[str, int, int]:
+    """Returns (group_name, min_available, headcount).
 
     The scheduler vocabulary derives min_available =
     floor(headcount*threshold + 0.5) (reference pod_group.go:86-117);
     the aggregator vocabulary is a direct sharedgpu/min_available label
-    (aggregator/pod.go:22,92). Both accepted; direct label wins.
+    (aggregator/pod.go:22,92). Both accepted; direct label wins for
+    min_available. headcount is kept separately because the Permit gang
+    timeout is 2 s x HEADCOUNT, not min_available (reference
+    scheduler.go:44,573); 0 when the label is absent.
     """
     group = labels.get(C.POD_GROUP_NAME, "")
+    headcount = 0
+    head_raw = labels.get(C.POD_GROUP_HEADCOUNT)
+    if head_raw is not None:
+        try:
+            headcount = int(head_raw)
+        except ValueError as e:
+            raise LabelError("group_headcount set error") from e
+        if headcount < 0:
+            raise LabelError("group_headcount negative")
     direct = labels.get(C.POD_MIN_AVAILABLE)
     if direct is not None:
         try:
-            return group, max(0, int(direct))
+            return group, max(0, int(direct)), headcount
         except ValueError as e:
             raise LabelError(f"{C.POD_MIN_AVAILABLE} set error: {direct!r}") from e
-    head_raw = labels.get(C.POD_GROUP_HEADCOUNT)
     thr_raw = labels.get(C.POD_GROUP_THRESHOLD)
     if head_raw is None or thr_raw is None:
-        return group, 0
+        return group, 0, headcount
     try:
-        headcount = int(head_raw)
         threshold = float(thr_raw)
     except ValueError as e:
         raise LabelError("group_headcount/group_threshold set error") from e
-    if headcount < 0 or not (0.0 <= threshold <= 1.0):
+    if not (0.0 <= threshold <= 1.0):
         raise LabelError("group_headcount/group_threshold out of range")
-    return group, int(math.floor(headcount * threshold + 0.5))
+    return group, int(math.floor(headcount * threshold + 0.5)), headcount
 
 
 def parse_pod(namespace: str, name: str, labels: dict, *, uid: str = "",
@@ -132,7 +143,7 @@ def parse_pod(namespace: str, name: str, labels: dict, *, uid: str = "",
     as regular).
     """
     spec = PodSpec(namespace=namespace, name=name, uid=uid, node_name=node_name)
-    spec.pod_group, spec.min_available = parse_gang(labels)
+    spec.pod_group, spec.min_available, spec.headcount = parse_gang(labels)
     spec.priority = parse_priority(labels)
 
     raw_limit = labels.get(C.POD_GPU_LIMIT)

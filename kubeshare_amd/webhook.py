@@ -24,9 +24,52 @@ import json
 from .utils import constants as C
 
 
+def shared_pod_env(namespace: str, name: str, port: str,
+                   gpu_mem: str, index: str) -> list:
+    """The isolation env block for one shared-GPU container (ROCm-native
+    equivalent of the reference's injection, pod.go:445-457). Transport:
+    UDS through the /kubeshare/sock hostPath by default; POD_MANAGER_IP
+    resolves to the NODE's IP via the downward API (the pod-mgr runs in
+    the hostNetwork node daemon — 127.0.0.1 inside a pod-network
+    container is the container itself, not the node)."""
+    return [
+        {"name": C.ENV_INJECTED, "value": "1"},
+        {"name": C.ENV_ROCR_VISIBLE_DEVICES, "value": index},
+        {"name": C.ENV_LD_PRELOAD, "value": C.HOOK_SO_PATH},
+        {"name": C.ENV_POD_MANAGER_UDS,
+         "value": C.pod_manager_uds(int(port))},
+        {"name": C.ENV_POD_MANAGER_IP,
+         "valueFrom": {"fieldRef": {"fieldPath": "status.hostIP"}}},
+        {"name": C.ENV_POD_MANAGER_PORT, "value": port},
+        {"name": C.ENV_POD_NAME, "value": f"{namespace}/{name}"},
+        {"name": C.ENV_GPU_MEM, "value": gpu_mem},
+        {"name": C.ENV_REQUIRE_HOOK, "value": "1"},
+    ]
+
+
+# volumes a shared-GPU pod needs: the hook .so (read-only) and the
+# pod-mgr socket dir (RW — connect() needs write access to the inode)
+SHARED_MOUNTS = [
+    {"name": "kubeshare-library", "mountPath": C.LIBRARY_PATH,
+     "readOnly": True},
+    {"name": "kubeshare-sock", "mountPath": C.SOCK_DIR},
+]
+SHARED_VOLUMES = [
+    {"name": "kubeshare-library", "hostPath": {"path": C.LIBRARY_PATH}},
+    {"name": "kubeshare-sock",
+     "hostPath": {"path": C.SOCK_DIR, "type": "DirectoryOrCreate"}},
+]
+
+
 def build_patch(pod: dict) -> list:
     """JSONPatch for one pod dict; [] when not a shared-GPU pod or
-    already injected."""
+    already injected.
+
+    Idempotency is keyed on the KUBESHARE_INJECTED marker — NOT on the
+    presence of ROCR_VISIBLE_DEVICES: a user-set device-visibility env
+    must not suppress injection (that would schedule the pod onto a
+    shared GPU with no hook, no pinning, no memory cap). Conflicting
+    user-set ROCR/HIP_VISIBLE_DEVICES entries are replaced in place."""
     meta = pod.get("metadata", {})
     ann = meta.get("annotations") or {}
     uuid = ann.get(C.POD_GPU_UUID)
@@ -37,54 +80,64 @@ def build_patch(pod: dict) -> list:
     index = ann.get(C.POD_GPU_INDEX, "")
     shared = bool(port)  # whole-GPU pods get no isolation layer
 
-    env = [{"name": C.ENV_ROCR_VISIBLE_DEVICES,
-            "value": index if index else uuid}]
     if shared:
-        env += [
-            {"name": C.ENV_LD_PRELOAD, "value": C.HOOK_SO_PATH},
-            {"name": C.ENV_POD_MANAGER_IP, "value": "127.0.0.1"},
-            {"name": C.ENV_POD_MANAGER_PORT, "value": port},
-            {"name": C.ENV_POD_NAME,
-             "value": f"{meta.get('namespace', 'default')}/{meta.get('name')}"},
-            {"name": C.ENV_GPU_MEM, "value": gpu_mem},
-            {"name": C.ENV_REQUIRE_HOOK, "value": "1"},
-        ]
+        env = shared_pod_env(meta.get("namespace", "default"),
+                             meta.get("name", ""), port, gpu_mem, index)
+    else:
+        env = [{"name": C.ENV_INJECTED, "value": "1"},
+               {"name": C.ENV_ROCR_VISIBLE_DEVICES,
+                "value": index if index else uuid}]
 
+    ours = {e["name"] for e in env} | {C.ENV_HIP_VISIBLE_DEVICES}
     patch = []
+    injected_any = False
     containers = pod.get("spec", {}).get("containers", [])
     for i, container in enumerate(containers):
-        existing = {e.get("name") for e in container.get("env") or []}
-        if C.ENV_ROCR_VISIBLE_DEVICES in existing:
-            continue  # already injected
-        if container.get("env") is None:
+        existing = container.get("env") or []
+        if any(e.get("name") == C.ENV_INJECTED for e in existing):
+            continue  # already injected (webhook re-invocation)
+        injected_any = True
+        # replace conflicting user-set entries in place, append the rest
+        conflict_idx = {e.get("name"): j for j, e in enumerate(existing)
+                        if e.get("name") in ours}
+        if not existing:
             patch.append({"op": "add", "path": f"/spec/containers/{i}/env",
                           "value": env})
         else:
             for e in env:
-                patch.append({"op": "add",
-                              "path": f"/spec/containers/{i}/env/-",
-                              "value": e})
+                j = conflict_idx.get(e["name"])
+                if j is not None:
+                    patch.append({"op": "replace",
+                                  "path": f"/spec/containers/{i}/env/{j}",
+                                  "value": e})
+                else:
+                    patch.append({"op": "add",
+                                  "path": f"/spec/containers/{i}/env/-",
+                                  "value": e})
         if shared:
-            mount = {"name": "kubeshare-library",
-                     "mountPath": C.LIBRARY_PATH, "readOnly": True}
-            if container.get("volumeMounts") is None:
+            mounts = container.get("volumeMounts")
+            have = {m.get("name") for m in mounts or []}
+            want = [m for m in SHARED_MOUNTS if m["name"] not in have]
+            if mounts is None:
                 patch.append({"op": "add",
                               "path": f"/spec/containers/{i}/volumeMounts",
-                              "value": [mount]})
+                              "value": want})
             else:
-                patch.append({"op": "add",
-                              "path": f"/spec/containers/{i}/volumeMounts/-",
-                              "value": mount})
-    if shared and patch:
-        vol = {"name": "kubeshare-library",
-               "hostPath": {"path": C.LIBRARY_PATH}}
+                for m in want:
+                    patch.append({"op": "add",
+                                  "path": f"/spec/containers/{i}/volumeMounts/-",
+                                  "value": m})
+    if shared and injected_any:
         vols = pod.get("spec", {}).get("volumes")
+        have = {v.get("name") for v in vols or []}
+        want = [v for v in SHARED_VOLUMES if v["name"] not in have]
         if vols is None:
             patch.append({"op": "add", "path": "/spec/volumes",
-                          "value": [vol]})
-        elif not any(v.get("name") == "kubeshare-library" for v in vols):
-            patch.append({"op": "add", "path": "/spec/volumes/-",
-                          "value": vol})
+                          "value": want})
+        else:
+            for v in want:
+                patch.append({"op": "add", "path": "/spec/volumes/-",
+                              "value": v})
     return patch
 
 

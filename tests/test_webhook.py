@@ -23,7 +23,7 @@ def test_patch_for_shared_pod():
     patch = build_patch(_pod_with_annotations(ann))
     env_ops = [p for p in patch if p["path"].endswith("/env")]
     assert env_ops, patch
-    env = {e["name"]: e["value"] for e in env_ops[0]["value"]}
+    env = {e["name"]: e.get("value") for e in env_ops[0]["value"]}
     assert env[C.ENV_ROCR_VISIBLE_DEVICES] == "3"
     assert env[C.ENV_LD_PRELOAD] == C.HOOK_SO_PATH
     assert env[C.ENV_POD_MANAGER_PORT] == "50050"
@@ -35,7 +35,7 @@ def test_patch_for_shared_pod():
 def test_patch_for_whole_gpu_pod_no_hook():
     ann = {C.POD_GPU_UUID: "GPU-1,GPU-2", C.POD_GPU_INDEX: "1,2"}
     patch = build_patch(_pod_with_annotations(ann))
-    env = {e["name"]: e["value"] for p in patch
+    env = {e["name"]: e.get("value") for p in patch
            if p["path"].endswith("/env") for e in p["value"]}
     assert env[C.ENV_ROCR_VISIBLE_DEVICES] == "1,2"
     assert C.ENV_LD_PRELOAD not in env
@@ -47,12 +47,34 @@ def test_patch_noop_for_regular_pod():
 
 
 def test_patch_idempotent():
+    """Idempotency keys on the KUBESHARE_INJECTED marker, not on
+    ROCR_VISIBLE_DEVICES (round-1 advisor finding: a user-set
+    device-visibility env must not silently bypass the hook)."""
     ann = {C.POD_GPU_UUID: "GPU-1", C.POD_MANAGER_PORT: "50050",
            C.POD_GPU_INDEX: "0"}
     pod = _pod_with_annotations(ann)
     pod["spec"]["containers"][0]["env"] = [
-        {"name": C.ENV_ROCR_VISIBLE_DEVICES, "value": "0"}]
+        {"name": C.ENV_INJECTED, "value": "1"}]
     assert build_patch(pod) == []
+
+
+def test_patch_replaces_user_set_visibility_env():
+    """A container that sets ROCR_VISIBLE_DEVICES itself still gets the
+    full injection — the conflicting entry is replaced in place."""
+    ann = {C.POD_GPU_UUID: "GPU-1", C.POD_MANAGER_PORT: "50050",
+           C.POD_GPU_MEMORY: "1000", C.POD_GPU_INDEX: "3"}
+    pod = _pod_with_annotations(ann)
+    pod["spec"]["containers"][0]["env"] = [
+        {"name": "FOO", "value": "bar"},
+        {"name": C.ENV_ROCR_VISIBLE_DEVICES, "value": "0,1,2,3"}]
+    patch = build_patch(pod)
+    replaces = [p for p in patch if p["op"] == "replace"]
+    assert any(p["path"] == "/spec/containers/0/env/1" and
+               p["value"]["value"] == "3" for p in replaces)
+    env_names = [p["value"]["name"] for p in patch
+                 if p["op"] == "add" and "/env/-" in p["path"]]
+    assert C.ENV_LD_PRELOAD in env_names
+    assert C.ENV_POD_MANAGER_UDS in env_names
 
 
 def test_admission_review_roundtrip():
@@ -80,7 +102,7 @@ def test_scheduler_annotations_feed_webhook():
                      "annotations": pod.annotations},
         "spec": {"containers": [{"name": "main"}]},
     })
-    env = {e["name"]: e["value"] for p in patch
+    env = {e["name"]: e.get("value") for p in patch
            if p["path"].endswith("/env") for e in p["value"]}
     assert env[C.ENV_POD_MANAGER_PORT] == \
         pod.annotations[C.POD_MANAGER_PORT]
