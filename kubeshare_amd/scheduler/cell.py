@@ -35,6 +35,9 @@ class Cell:
     free_memory: int = 0
     full_memory: int = 0
     uuid: str = ""             # leaf only
+    # leaf only: peer-uuid -> xGMI link count (None = unknown topology,
+    # assume the MI355X 7-link clique; missing peer = link down/absent)
+    xgmi_peers: Optional[dict] = None
 
     @property
     def is_leaf(self) -> bool:
@@ -60,11 +63,14 @@ class CellTree:
         # model -> list of root cells whose leaf type is that model
         self.roots_by_model: dict[str, list[Cell]] = {}
         self.leaf_by_uuid: dict[str, Cell] = {}
+        self.leaf_by_id: dict[str, Cell] = {}
         self.node_cells: dict[str, list[Cell]] = {}  # node -> node-level cells
         for spec in cfg.cells:
             root = self._build(spec, parent=None)
             self.roots_by_model.setdefault(
                 self.elements[root.cell_type].leaf_cell_type, []).append(root)
+            for leaf in root.leaves():
+                self.leaf_by_id[leaf.id] = leaf
         # models sorted by priority desc (reference sortGPUPriority
         # cell.go:57-72) — scheduling considers faster models first
         self.models_by_priority = sorted(
@@ -110,10 +116,19 @@ class CellTree:
             model = self.elements[node_cell.cell_type].leaf_cell_type
             gpus = list(gpus_by_model.get(model, []))
             leaves = [c for c in node_cell.leaves()]
+            # node-local GPU index -> uuid, to translate the inventory's
+            # index-keyed xGMI adjacency into uuid-keyed leaf peers
+            idx2uuid = {g.get("index", i): g["uuid"]
+                        for i, g in enumerate(gpus)}
             for leaf, gpu in zip(leaves, gpus):
                 first_fill = not leaf.uuid
                 leaf.uuid = gpu["uuid"]
                 leaf.full_memory = int(gpu["memory"])
+                links = gpu.get("xgmi_links")
+                if links is not None:
+                    leaf.xgmi_peers = {
+                        idx2uuid[j]: w for j, w in links.items()
+                        if j in idx2uuid and w > 0}
                 if first_fill:
                     leaf.free_memory = leaf.full_memory
                     leaf.available = 1.0

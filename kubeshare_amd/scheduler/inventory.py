@@ -44,10 +44,19 @@ class FakeInventory:
 
     def add_node(self, node: str, gpus: int = 8,
                  model: str = C.MI355X_MODEL,
-                 memory: int = C.MI355X_HBM_BYTES):
+                 memory: int = C.MI355X_HBM_BYTES,
+                 down_links: list | None = None):
+        """`down_links`: list of (i, j) GPU-index pairs whose direct
+        xGMI link is absent/failed (removed from both directions) — lets
+        tests model a degraded topology."""
+        down = set()
+        for i, j in down_links or []:
+            down.add((i, j))
+            down.add((j, i))
         infos = []
         for i in range(gpus):
-            links = {j: 1 for j in range(gpus) if j != i}
+            links = {j: 1 for j in range(gpus)
+                     if j != i and (i, j) not in down}
             infos.append(GPUInfo(uuid=f"GPU-{node}-{i}", model=model,
                                  memory=memory, index=i, xgmi_links=links))
         self._nodes[node] = infos
@@ -59,8 +68,23 @@ class FakeInventory:
         out: dict[str, list] = {}
         for g in self.gpus(node):
             out.setdefault(g.model, []).append(
-                {"uuid": g.uuid, "memory": g.memory, "index": g.index})
+                {"uuid": g.uuid, "memory": g.memory, "index": g.index,
+                 "xgmi_links": dict(g.xgmi_links)})
         return out
+
+
+def format_node_annotation(gpus: list) -> str:
+    """Serialize local inventory (list[GPUInfo]) into the
+    `kubeshare.amd/gpus` node annotation the KubeDriver consumes:
+    "uuid,model,memory,index,links=j:k;..." — links are the peer GPU
+    indices with a live direct xGMI link."""
+    parts = []
+    for g in gpus:
+        links = ":".join(str(j) for j, w in sorted(g.xgmi_links.items())
+                         if w > 0)
+        parts.append(f"{g.uuid},{g.model},{g.memory},{g.index},"
+                     f"links={links}")
+    return ";".join(parts)
 
 
 class AmdSmiInventory:

@@ -70,15 +70,25 @@ class KubeDriver:
                 by_model = self.inventory.by_model(name)
             else:
                 # collector publishes inventory as node annotations
-                # kubeshare.amd/gpus = "uuid,model,memory,index;..."
+                # kubeshare.amd/gpus =
+                #   "uuid,model,memory,index[,links=j:k:l];..."
+                # (links = peer GPU indices with a live direct xGMI
+                # link; absent -> assume the MI355X clique)
                 raw = (node.metadata.annotations or {}).get(
                     "kubeshare.amd/gpus", "")
                 by_model = {}
                 for entry in filter(None, raw.split(";")):
-                    uuid, model, memory, index = entry.split(",")
-                    by_model.setdefault(model, []).append(
-                        {"uuid": uuid, "memory": int(memory),
-                         "index": int(index)})
+                    fields = entry.split(",")
+                    uuid, model, memory, index = fields[:4]
+                    gpu = {"uuid": uuid, "memory": int(memory),
+                           "index": int(index)}
+                    for extra in fields[4:]:
+                        if extra.startswith("links="):
+                            gpu["xgmi_links"] = {
+                                int(j): 1
+                                for j in extra[len("links="):].split(":")
+                                if j}
+                    by_model.setdefault(model, []).append(gpu)
             ready = any(c.type == "Ready" and c.status == "True"
                         for c in (node.status.conditions or []))
             if by_model:

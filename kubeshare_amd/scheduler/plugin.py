@@ -204,11 +204,26 @@ class KubeShareScheduler:
                 for cid in s.cell_ids]
 
     def _distance(self, cell: Cell, other_id: str) -> float:
-        """Locality distance. xGMI-aware at leaf level: within one
-        MI355X node every GPU pair is one direct link (7-link clique),
-        so intra-node distance is uniform and small; across nodes the
-        hierarchical digit distance dominates (reference
-        getCellIDDistance score.go:164-227)."""
+        """Locality distance for gang placement.
+
+        Intra-node the distance comes from the REAL xGMI link graph read
+        from amdsmi (inventory.GPUInfo.xgmi_links -> Cell.xgmi_peers):
+        0 = same GPU, 1 = direct xGMI link, 2 = link down/absent (the
+        hop must route through a third GPU or host). On a healthy MI355X
+        this degenerates to the 7-link clique (every pair = 1), but a
+        degraded link now visibly demotes that pair in Score/Reserve.
+        When adjacency is unknown (no amdsmi, plain YAML topology) the
+        clique is assumed. Across nodes: the reference's hierarchical
+        cell-ID digit distance (getCellIDDistance score.go:164-227)."""
+        other = self.tree.leaf_by_id.get(other_id)
+        if other is not None and cell.node_name and \
+                other.node_name == cell.node_name:
+            if cell.id == other_id:
+                return 0.0
+            peers = cell.xgmi_peers
+            if peers is None or not other.uuid:
+                return 1.0  # unknown topology: assume the clique
+            return 1.0 if peers.get(other.uuid, 0) > 0 else 2.0
         a = cell.id.split("/")
         b = other_id.split("/")
         dist = 0.0
@@ -217,17 +232,17 @@ class KubeShareScheduler:
             ca = a[la - k] if k <= la else None
             cb = b[lb - k] if k <= lb else None
             if ca is None or cb is None:
-                other = cb if ca is None else ca
+                o = cb if ca is None else ca
                 try:
-                    dist += abs(int(other))
+                    dist += abs(int(o))
                 except (TypeError, ValueError):
                     dist += 100.0
                 continue
             try:
                 ia, ib = int(ca), int(cb)
                 if k == 1:
-                    # leaf position inside the node: one xGMI hop apart
-                    # regardless of index on MI355X's clique
+                    # leaf digit across nodes: position is meaningless,
+                    # any pair is "one device apart"
                     dist += 0.0 if ia == ib else 1.0
                 else:
                     dist += abs(ia - ib)

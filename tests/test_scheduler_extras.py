@@ -166,3 +166,99 @@ def test_graft_entry_contract():
     mod = importlib.util.module_from_spec(spec)
     spec.loader.exec_module(mod)
     assert callable(mod.build) and callable(mod.smoke)
+
+
+# ---------------------------------------------------------------- xGMI score
+class TestXgmiAwarePlacement:
+    """Score/Reserve consume the real xGMI link graph (GPUInfo.xgmi_links
+    -> Cell.xgmi_peers): a degraded link demotes that GPU pair for gang
+    placement; a healthy MI355X degenerates to the 7-link clique.
+    Replaces the reference's pure string heuristic (score.go:164-227)."""
+
+    def _cluster(self, down_links):
+        from kubeshare_amd.scheduler.harness import FakeCluster
+        from kubeshare_amd.scheduler.inventory import FakeInventory
+        fc = FakeCluster(nodes={"node-a": {"gpus": 4}})
+        inv = FakeInventory()
+        inv.add_node("node-a", gpus=4, down_links=down_links)
+        fc.inventory = inv
+        fc.scheduler.register_node("node-a", inv.by_model("node-a"))
+        return fc
+
+    def _gang_pod(self, fc, name):
+        # full-GPU shares: each rank needs its own GPU, so the xGMI
+        # locality term decides WHICH one (a 0.5+0.5 gang would instead
+        # co-locate on one GPU — distance 0 beats any link)
+        return fc.add_pod("ns", name, {
+            C.POD_GPU_REQUEST: "1.0", C.POD_GPU_LIMIT: "1.0",
+            C.POD_PRIORITY: "100", C.POD_GROUP_NAME: "g",
+            C.POD_GROUP_HEADCOUNT: "2", C.POD_GROUP_THRESHOLD: "1.0"})
+
+    def test_healthy_clique_packs_anywhere(self):
+        fc = self._cluster(down_links=[])
+        self._gang_pod(fc, "r0")
+        self._gang_pod(fc, "r1")
+        fc.schedule_pending()
+        uuids = {fc.pods[f"ns/r{i}"].annotations[C.POD_GPU_UUID]
+                 for i in (0, 1)}
+        assert len(uuids) == 2  # both bound, each on a GPU
+
+    def test_degraded_link_changes_placement(self):
+        # rank0 lands on GPU-0; links 0-1 and 0-2 are down, so rank1
+        # must prefer GPU-3 (direct link) over GPU-1/2 (2 hops)
+        fc = self._cluster(down_links=[(0, 1), (0, 2)])
+        self._gang_pod(fc, "r0")
+        self._gang_pod(fc, "r1")
+        fc.schedule_pending()
+        first = fc.pods["ns/r0"].annotations[C.POD_GPU_UUID]
+        assert first == "GPU-node-a-0"
+        second = fc.pods["ns/r1"].annotations[C.POD_GPU_UUID]
+        assert second == "GPU-node-a-3"
+
+    def test_distance_values(self):
+        fc = self._cluster(down_links=[(0, 1)])
+        sch = fc.scheduler
+        leaf0 = sch.tree.leaf_by_uuid["GPU-node-a-0"]
+        leaf1 = sch.tree.leaf_by_uuid["GPU-node-a-1"]
+        leaf2 = sch.tree.leaf_by_uuid["GPU-node-a-2"]
+        assert sch._distance(leaf0, leaf0.id) == 0.0
+        assert sch._distance(leaf0, leaf2.id) == 1.0  # direct link
+        assert sch._distance(leaf0, leaf1.id) == 2.0  # link down
+        assert sch._distance(leaf1, leaf0.id) == 2.0  # symmetric
+
+    def test_unknown_topology_assumes_clique(self):
+        from kubeshare_amd.scheduler.harness import FakeCluster
+        fc = FakeCluster(nodes={"node-a": {"gpus": 4}})
+        sch = fc.scheduler
+        leaf0 = sch.tree.leaf_by_uuid["GPU-node-a-0"]
+        leaf1 = sch.tree.leaf_by_uuid["GPU-node-a-1"]
+        # FakeCluster default inventory is a full clique
+        assert sch._distance(leaf0, leaf1.id) == 1.0
+
+
+def test_node_annotation_roundtrip_with_links():
+    """format_node_annotation -> KubeDriver.sync_nodes parse preserves
+    the link graph (degraded pair ends up non-adjacent in the tree)."""
+    from types import SimpleNamespace as NS
+    from kubeshare_amd.scheduler.inventory import (FakeInventory,
+                                                   format_node_annotation)
+    from kubeshare_amd.scheduler.kube import KubeDriver
+    from kubeshare_amd.scheduler.topology import TopologyConfig
+    inv = FakeInventory()
+    inv.add_node("node-a", gpus=4, down_links=[(1, 2)])
+    ann = format_node_annotation(inv.gpus("node-a"))
+
+    class V1:
+        def list_node(self, label_selector=None):
+            return NS(items=[NS(
+                metadata=NS(name="node-a",
+                            annotations={"kubeshare.amd/gpus": ann}),
+                status=NS(conditions=[NS(type="Ready", status="True")]))])
+    d = KubeDriver(TopologyConfig.single_node("node-a", gpus=4), api=V1())
+    d.sync_nodes()
+    l1 = d.sched.tree.leaf_by_uuid["GPU-node-a-1"]
+    l2 = d.sched.tree.leaf_by_uuid["GPU-node-a-2"]
+    l3 = d.sched.tree.leaf_by_uuid["GPU-node-a-3"]
+    assert l3.uuid in l1.xgmi_peers
+    assert l2.uuid not in l1.xgmi_peers
+    assert d.sched._distance(l1, l2.id) == 2.0
