@@ -155,6 +155,24 @@ HookState& S() {
     } else if (sip && sip[0] && sport && sport[0]) {
       host = sip;
       p = atoi(sport);
+    } else if (port && port[0]) {
+      // Port injected but no IP: find the node-local manager the way
+      // the reference's gemhook does — schedulerIP.txt on the mounted
+      // /kubeshare/library hostPath (written by kubeshare-query-ip,
+      // reference cmd/kubeshare-query-ip/main.go:23-34).
+      const char* ipfile = getenv("KUBESHARE_SCHEDULER_IP_FILE");
+      if (!ipfile) ipfile = "/kubeshare/library/schedulerIP.txt";
+      if (FILE* f = fopen(ipfile, "r")) {
+        char buf[128] = {0};
+        if (fgets(buf, sizeof(buf), f)) {
+          buf[strcspn(buf, " \t\r\n")] = 0;
+          if (buf[0]) {
+            host = buf;
+            p = atoi(port);
+          }
+        }
+        fclose(f);
+      }
     }
     if (!st->disabled && !host.empty()) {
       st->gate.init(
@@ -227,6 +245,10 @@ inline void gate2(void* ra, hipStream_t stream) {
 }
 
 // ------------------------------------------------------------ mem tracking
+// Reserve-then-commit: try_reserve BUMPS `used` (so concurrent
+// allocations can't all pass the check and overshoot the cap — round-1
+// advisor finding); commit records the pointer, unreserve rolls back
+// when the real allocation failed.
 bool mem_try_reserve(size_t size) {
   HookState& s = S();
   if (s.cap == 0) return true;
@@ -235,14 +257,35 @@ bool mem_try_reserve(size_t size) {
     s.denied++;
     return false;
   }
+  s.used += size;
   return true;
+}
+
+void mem_unreserve(size_t size) {
+  HookState& s = S();
+  if (s.cap == 0) return;
+  std::lock_guard<std::mutex> g(s.mem_mu);
+  s.used = s.used >= size ? s.used - size : 0;
 }
 
 void mem_commit(void* ptr, size_t size) {
   HookState& s = S();
   std::lock_guard<std::mutex> g(s.mem_mu);
   s.allocs[ptr] = size;
-  s.used += size;
+  if (s.cap == 0) s.used += size;  // cap set: already counted at reserve
+}
+
+// Reserved `reserved` bytes, actual allocation came out `actual` (pitch
+// padding): settle the difference.
+void mem_commit_adjusted(void* ptr, size_t reserved, size_t actual) {
+  HookState& s = S();
+  std::lock_guard<std::mutex> g(s.mem_mu);
+  s.allocs[ptr] = actual;
+  if (s.cap == 0) {
+    s.used += actual;
+  } else {
+    s.used = s.used + actual - reserved;
+  }
 }
 
 void mem_release(void* ptr) {
@@ -267,6 +310,7 @@ hipError_t hipMalloc(void** ptr, size_t size) {
   if (!mem_try_reserve(size)) return hipErrorOutOfMemory;
   hipError_t e = real(ptr, size);
   if (e == hipSuccess && ptr && *ptr) mem_commit(*ptr, size);
+  else mem_unreserve(size);
   return e;
 }
 
@@ -276,6 +320,7 @@ hipError_t hipExtMallocWithFlags(void** ptr, size_t size, unsigned int flags) {
   if (!mem_try_reserve(size)) return hipErrorOutOfMemory;
   hipError_t e = real(ptr, size, flags);
   if (e == hipSuccess && ptr && *ptr) mem_commit(*ptr, size);
+  else mem_unreserve(size);
   return e;
 }
 
@@ -285,6 +330,7 @@ hipError_t hipMallocManaged(void** ptr, size_t size, unsigned int flags) {
   if (!mem_try_reserve(size)) return hipErrorOutOfMemory;
   hipError_t e = real(ptr, size, flags);
   if (e == hipSuccess && ptr && *ptr) mem_commit(*ptr, size);
+  else mem_unreserve(size);
   return e;
 }
 
@@ -294,6 +340,7 @@ hipError_t hipMallocAsync(void** ptr, size_t size, hipStream_t stream) {
   if (!mem_try_reserve(size)) return hipErrorOutOfMemory;
   hipError_t e = real(ptr, size, stream);
   if (e == hipSuccess && ptr && *ptr) mem_commit(*ptr, size);
+  else mem_unreserve(size);
   return e;
 }
 
@@ -304,6 +351,7 @@ hipError_t hipMallocFromPoolAsync(void** ptr, size_t size, hipMemPool_t pool,
   if (!mem_try_reserve(size)) return hipErrorOutOfMemory;
   hipError_t e = real(ptr, size, pool, stream);
   if (e == hipSuccess && ptr && *ptr) mem_commit(*ptr, size);
+  else mem_unreserve(size);
   return e;
 }
 
@@ -334,6 +382,131 @@ hipError_t hipMemGetInfo(size_t* free_b, size_t* total_b) {
     if (total_b && *total_b > s.cap) *total_b = s.cap;
     if (free_b && *free_b > cap_free) *free_b = cap_free;
   }
+  return e;
+}
+
+// ----------------------------------------------- VMM / pitched / arrays
+// Round-1 gap (VERDICT "Missing #5"): PyTorch's expandable_segments
+// allocator goes through hipMemCreate/hipMemMap, and pitched/3D/array
+// allocations have their own entry points — all bypassed the
+// KUBESHARE_GPU_MEM cap (reference contract: gpu_mem default
+// floor(request x fullMemory), pod.go:419-421). Physical memory is
+// committed at hipMemCreate (hipMemAddressReserve is VA only, hipMemMap
+// maps an existing handle), so that is where the cap applies.
+
+hipError_t hipMemCreate(hipMemGenericAllocationHandle_t* handle, size_t size,
+                        const hipMemAllocationProp* prop,
+                        unsigned long long flags) {
+  REAL(hipMemCreate, hipMemGenericAllocationHandle_t*, size_t,
+       const hipMemAllocationProp*, unsigned long long);
+  if (!real) return hipErrorNotInitialized;
+  if (!mem_try_reserve(size)) return hipErrorOutOfMemory;
+  hipError_t e = real(handle, size, prop, flags);
+  if (e == hipSuccess && handle && *handle)
+    mem_commit((void*)*handle, size);
+  else
+    mem_unreserve(size);
+  return e;
+}
+
+hipError_t hipMemRelease(hipMemGenericAllocationHandle_t handle) {
+  REAL(hipMemRelease, hipMemGenericAllocationHandle_t);
+  if (!real) return hipErrorNotInitialized;
+  hipError_t e = real(handle);
+  if (e == hipSuccess && handle) mem_release((void*)handle);
+  return e;
+}
+
+hipError_t hipMallocPitch(void** ptr, size_t* pitch, size_t width,
+                          size_t height) {
+  REAL(hipMallocPitch, void**, size_t*, size_t, size_t);
+  if (!real) return hipErrorNotInitialized;
+  size_t est = width * height;  // lower bound; pitch padding settled below
+  if (!mem_try_reserve(est)) return hipErrorOutOfMemory;
+  hipError_t e = real(ptr, pitch, width, height);
+  if (e == hipSuccess && ptr && *ptr && pitch)
+    mem_commit_adjusted(*ptr, est, *pitch * height);
+  else
+    mem_unreserve(est);
+  return e;
+}
+
+hipError_t hipMemAllocPitch(hipDeviceptr_t* dptr, size_t* pitch,
+                            size_t widthInBytes, size_t height,
+                            unsigned int elementSizeBytes) {
+  REAL(hipMemAllocPitch, hipDeviceptr_t*, size_t*, size_t, size_t,
+       unsigned int);
+  if (!real) return hipErrorNotInitialized;
+  size_t est = widthInBytes * height;
+  if (!mem_try_reserve(est)) return hipErrorOutOfMemory;
+  hipError_t e = real(dptr, pitch, widthInBytes, height, elementSizeBytes);
+  if (e == hipSuccess && dptr && *dptr && pitch)
+    mem_commit_adjusted((void*)*dptr, est, *pitch * height);
+  else
+    mem_unreserve(est);
+  return e;
+}
+
+hipError_t hipMalloc3D(hipPitchedPtr* pitchedDevPtr, hipExtent extent) {
+  REAL(hipMalloc3D, hipPitchedPtr*, hipExtent);
+  if (!real) return hipErrorNotInitialized;
+  size_t est = extent.width * extent.height * extent.depth;
+  if (!mem_try_reserve(est)) return hipErrorOutOfMemory;
+  hipError_t e = real(pitchedDevPtr, extent);
+  if (e == hipSuccess && pitchedDevPtr && pitchedDevPtr->ptr)
+    mem_commit_adjusted(pitchedDevPtr->ptr, est,
+                        pitchedDevPtr->pitch * extent.height * extent.depth);
+  else
+    mem_unreserve(est);
+  return e;
+}
+
+static size_t array_bytes(const hipChannelFormatDesc* desc, size_t w,
+                          size_t h, size_t d) {
+  size_t texel = desc ? ((size_t)(desc->x + desc->y + desc->z + desc->w) + 7)
+                            / 8
+                      : 4;
+  if (h == 0) h = 1;
+  if (d == 0) d = 1;
+  return texel * w * h * d;
+}
+
+hipError_t hipMallocArray(hipArray_t* array, const hipChannelFormatDesc* desc,
+                          size_t width, size_t height, unsigned int flags) {
+  REAL(hipMallocArray, hipArray_t*, const hipChannelFormatDesc*, size_t,
+       size_t, unsigned int);
+  if (!real) return hipErrorNotInitialized;
+  size_t est = array_bytes(desc, width, height, 1);
+  if (!mem_try_reserve(est)) return hipErrorOutOfMemory;
+  hipError_t e = real(array, desc, width, height, flags);
+  if (e == hipSuccess && array && *array)
+    mem_commit((void*)*array, est);
+  else
+    mem_unreserve(est);
+  return e;
+}
+
+hipError_t hipMalloc3DArray(hipArray_t* array,
+                            const hipChannelFormatDesc* desc, hipExtent extent,
+                            unsigned int flags) {
+  REAL(hipMalloc3DArray, hipArray_t*, const hipChannelFormatDesc*, hipExtent,
+       unsigned int);
+  if (!real) return hipErrorNotInitialized;
+  size_t est = array_bytes(desc, extent.width, extent.height, extent.depth);
+  if (!mem_try_reserve(est)) return hipErrorOutOfMemory;
+  hipError_t e = real(array, desc, extent, flags);
+  if (e == hipSuccess && array && *array)
+    mem_commit((void*)*array, est);
+  else
+    mem_unreserve(est);
+  return e;
+}
+
+hipError_t hipFreeArray(hipArray_t array) {
+  REAL(hipFreeArray, hipArray_t);
+  if (!real) return hipErrorNotInitialized;
+  hipError_t e = real(array);
+  if (e == hipSuccess && array) mem_release((void*)array);
   return e;
 }
 

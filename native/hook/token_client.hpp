@@ -165,7 +165,7 @@ class TokenGate {
     pthread_mutex_lock(&mu_);
     if (holding_) {
       if (drain_) drain_(drain_arg_);
-      client_.return_token(now_ms() - grant_time_);
+      give_back(now_ms() - grant_time_);
       holding_ = false;
       deadline_.store(-1.0, std::memory_order_release);
     }
@@ -226,13 +226,22 @@ class TokenGate {
         } else {
           used = t_post - grant_time_;  // drain did real work
         }
-        client_.return_token(std::max(0.0, used));
+        give_back(std::max(0.0, used));
         holding_ = false;
         deadline_.store(-1.0, std::memory_order_release);
         idle_releases_++;
       }
       pthread_mutex_unlock(&mu_);
     }
+  }
+
+  // mu_ held. Reports use upstream and maintains the per-lease EWMA
+  // that renew() sends as the REQ hint.
+  void give_back(double used_ms) {
+    used_ms = std::max(0.0, used_ms);
+    ewma_used_ = ewma_used_ <= 0.0 ? used_ms
+                                   : 0.7 * ewma_used_ + 0.3 * used_ms;
+    client_.return_token(used_ms);
   }
 
   void start_watchdog_locked() {
@@ -256,11 +265,14 @@ class TokenGate {
       // submitted inside the lease.
       if (drain_) drain_(drain_arg_);
       double used = now_ms() - grant_time_;
-      client_.return_token(used);
+      give_back(used);
       holding_ = false;
     }
     start_watchdog_locked();
-    double quota = client_.acquire(0.0);
+    // hint = EWMA of recent per-lease use, so gpu-schd can right-size
+    // the next quota (a bursty pod gets shorter leases; see
+    // token_sched.hpp schedule())
+    double quota = client_.acquire(ewma_used_);
     if (quota <= 0.0) {
       // Scheduler unreachable: fail OPEN for liveness (the node daemon
       // restarts the chain; isolation degrades, jobs don't die) unless
@@ -289,6 +301,7 @@ class TokenGate {
   double idle_release_ms_ = 25.0;
   bool holding_ = false;
   double grant_time_ = 0.0;
+  double ewma_used_ = 0.0;
 
  public:
   long long failures_ = 0;

@@ -19,6 +19,7 @@
 #include <memory>
 #include <set>
 
+#include "busy_sampler.hpp"
 #include "token_sched.hpp"
 
 using namespace ks;
@@ -27,9 +28,9 @@ namespace {
 
 struct Client {
   int fd;
+  long long conn_id;            // unique (fds are reused by the kernel)
   LineBuffer rx;
   std::set<long long> pending;  // outstanding REQ cookies of this conn
-  std::string last_pod;
 };
 
 FILE* g_log = stderr;
@@ -39,10 +40,11 @@ FILE* g_log = stderr;
 int main(int argc, char** argv) {
   std::string config_dir, config_file, uds_path, log_path;
   int port = 0;
+  int gpu_index = -1;  // -d: enable the server-side busy sampler
   double base_q = 300.0, min_q = 20.0, window = 10000.0;
 
   int opt;
-  while ((opt = getopt(argc, argv, "p:f:P:q:m:w:U:l:")) != -1) {
+  while ((opt = getopt(argc, argv, "p:f:P:q:m:w:U:l:d:")) != -1) {
     switch (opt) {
       case 'p': config_dir = optarg; break;
       case 'f': config_file = optarg; break;
@@ -52,10 +54,11 @@ int main(int argc, char** argv) {
       case 'w': window = atof(optarg); break;
       case 'U': uds_path = optarg; break;
       case 'l': log_path = optarg; break;
+      case 'd': gpu_index = atoi(optarg); break;
       default:
         fprintf(stderr,
                 "usage: gpu-schd -p <dir> -f <file> -P <port> [-q ms] [-m ms] "
-                "[-w ms] [-U uds] [-l log]\n");
+                "[-w ms] [-U uds] [-l log] [-d gpu_index]\n");
         return 2;
     }
   }
@@ -70,6 +73,17 @@ int main(int argc, char** argv) {
   signal(SIGPIPE, SIG_IGN);
 
   TokenScheduler sched(base_q, min_q, window);
+
+  BusySampler sampler;
+  if (gpu_index >= 0) {
+    if (sampler.init(gpu_index)) {
+      logf(g_log, "gpu-schd", "busy sampler on (gpu %d): leases charged "
+           "sampled GPU-busy, not wall", gpu_index);
+    } else {
+      logf(g_log, "gpu-schd", "busy sampler unavailable (gpu %d): "
+           "falling back to wall/RET accounting", gpu_index);
+    }
+  }
 
   // Order matters: install the config WATCH before anyone can observe
   // us as ready (the listen socket), then load — a config written
@@ -104,6 +118,7 @@ int main(int argc, char** argv) {
   std::map<int, std::unique_ptr<Client>> clients;
   std::map<long long, int> cookie2fd;  // outstanding REQ cookie -> client fd
   long long next_cookie = 1;
+  long long next_conn = 1;
   double wake_in = -1.0;  // ms until a capped waiter may become eligible
 
   auto flush_grants = [&]() {
@@ -122,7 +137,12 @@ int main(int argc, char** argv) {
         continue;
       }
       auto cl = clients.find(it->second);
-      if (cl != clients.end()) cl->second->pending.erase(g.cookie);
+      if (cl != clients.end()) {
+        cl->second->pending.erase(g.cookie);
+        // remember the owning CONNECTION so only its death frees the
+        // grant (a same-pod sibling connection dying must not)
+        sched.set_holder_owner(g.pod, cl->second->conn_id);
+      }
       char line[64];
       snprintf(line, sizeof(line), "GRANT %.3f", g.quota_ms);
       if (!send_line(it->second, line)) {
@@ -141,8 +161,12 @@ int main(int argc, char** argv) {
 
     int timeout = 200;  // liveness-check cadence
     if (wake_in > 0.0) timeout = (int)std::min(wake_in, 200.0) + 1;
+    // the sampler needs a short tick to attribute busy time accurately
+    if (sampler.active() && sched.n_holders() > 0 && timeout > 5)
+      timeout = 5;
     ::poll(pfds.data(), pfds.size(), timeout);
     wake_in = -1.0;
+    if (sampler.active()) sched.add_busy(sampler.poll(), now_ms());
 
     size_t idx = 0;
     // new connections
@@ -153,6 +177,7 @@ int main(int argc, char** argv) {
         setsockopt(cfd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
         auto c = std::make_unique<Client>();
         c->fd = cfd;
+        c->conn_id = next_conn++;
         clients[cfd] = std::move(c);
         if (clients.size() >= 1024) break;
       }
@@ -203,7 +228,6 @@ int main(int argc, char** argv) {
           double hint = tok.size() >= 3 ? atof(tok[2].c_str()) : 0.0;
           long long ck = next_cookie++;
           c.pending.insert(ck);
-          c.last_pod = tok[1];
           cookie2fd[ck] = c.fd;
           sched.request(tok[1], ck, hint, now);
         } else if (tok[0] == "RET" && tok.size() >= 3) {
@@ -221,11 +245,13 @@ int main(int argc, char** argv) {
     }
     for (int fd : dead) {
       Client& c = *clients[fd];
-      // cancel outstanding REQs from this conn, release its token
+      // cancel outstanding REQs from this conn; free exactly the
+      // grants THIS connection owns (not everything under its pod name
+      // — a live sibling connection may hold a granted token)
       sched.drop_waiters(c.pending);
       for (auto it = cookie2fd.begin(); it != cookie2fd.end();)
         it = (it->second == fd) ? cookie2fd.erase(it) : std::next(it);
-      if (!c.last_pod.empty()) sched.force_release(c.last_pod, now_ms());
+      sched.force_release_owned(c.conn_id, now_ms());
       ::close(fd);
       clients.erase(fd);
     }

@@ -255,6 +255,85 @@ int main() {
     CHECK(js.find("\"grants\":1") != std::string::npos);
   }
 
+  // ---- 13. grant-owner tracking: a dying sibling connection must not
+  // free a grant another connection holds (round-1 advisor finding)
+  {
+    TokenScheduler s(100, 20, 10000);
+    s.set_config({quota("a", 1.0, 0.5)});
+    Grant g;
+    double retry;
+    s.request("a", 1, 0, 0.0);
+    CHECK(s.schedule(0.0, &g, &retry));
+    s.set_holder_owner("a", 7);     // conn 7 produced the grant
+    s.force_release_owned(9, 10.0);  // conn 9 (same pod) dies
+    CHECK(s.n_holders() == 1);       // grant untouched
+    s.force_release_owned(7, 50.0);  // the owning conn dies
+    CHECK(s.n_holders() == 0);
+    CHECK(s.account("a").total_used_ms == 50.0);  // charged wall
+  }
+
+  // ---- 14. hint-sized quotas: EWMA hint right-sizes the lease
+  {
+    TokenScheduler s(300, 20, 10000);
+    s.set_config({quota("a", 1.0, 0.5)});
+    Grant g;
+    double retry;
+    s.request("a", 1, 50.0, 0.0);   // hint 50 -> quota 75 (x1.5)
+    CHECK(s.schedule(0.0, &g, &retry));
+    CHECK(g.quota_ms == 75.0);
+    s.release("a", 75, 75.0);
+    s.request("a", 2, 3.0, 75.0);   // tiny hint clamps to min_q
+    CHECK(s.schedule(75.0, &g, &retry));
+    CHECK(g.quota_ms == 20.0);
+    s.release("a", 20, 95.0);
+    s.request("a", 3, 0.0, 95.0);   // no hint -> base quota
+    CHECK(s.schedule(95.0, &g, &retry));
+    CHECK(g.quota_ms == 300.0);
+    s.release("a", 300, 395.0);
+    s.request("a", 4, 5000.0, 395.0);  // huge hint clamps to base_q
+    CHECK(s.schedule(395.0, &g, &retry));
+    CHECK(g.quota_ms == 300.0);
+  }
+
+  // ---- 15. server-side busy sampling: charge sampled GPU time, not
+  // wall — a CPU-bound phase inside the lease costs nothing
+  {
+    TokenScheduler s(300, 20, 10000);
+    s.set_config({quota("a", 1.0, 0.5)});
+    Grant g;
+    double retry;
+    s.request("a", 1, 0, 0.0);
+    CHECK(s.schedule(0.0, &g, &retry));
+    // 300 ms of wall, but the sampler saw only 90 ms of GPU activity
+    for (int i = 0; i < 30; i++) s.add_busy(3.0, i * 10.0);
+    s.release("a", 300.0, 300.0);  // client reports wall; ignored
+    CHECK(s.account("a").total_used_ms == 90.0);
+    // unattributed busy (no holder) is not charged to anyone
+    double before = s.account("a").total_used_ms;
+    s.add_busy(50.0, 400.0);
+    CHECK(s.account("a").total_used_ms == before);
+    CHECK(s.stats_json(400.0).find("\"sampler\":true") != std::string::npos);
+  }
+
+  // ---- 16. sampled busy splits evenly across gang co-holders
+  {
+    TokenScheduler s(300, 20, 10000);
+    s.set_config({quota("g1", 1.0, 0.5, "gang"), quota("g2", 1.0, 0.5,
+                                                       "gang")});
+    Grant g;
+    double retry;
+    s.request("g1", 1, 0, 0.0);
+    s.request("g2", 2, 0, 0.0);
+    CHECK(s.schedule(0.0, &g, &retry));
+    CHECK(s.schedule(0.0, &g, &retry));  // co-granted (same gang)
+    CHECK(s.n_holders() == 2);
+    s.add_busy(100.0, 50.0);
+    s.release("g1", 300.0, 300.0);
+    s.release("g2", 300.0, 300.0);
+    CHECK(s.account("g1").total_used_ms == 50.0);
+    CHECK(s.account("g2").total_used_ms == 50.0);
+  }
+
   printf("sched_test OK (%d checks)\n", g_checks);
   return 0;
 }

@@ -120,14 +120,24 @@ class TokenScheduler {
 
   // A holder returned its token. `used_ms` is the client-measured GPU
   // time; it is clamped server-side to [0, wall + slack] so a lying
-  // client cannot under-report what it actually occupied.
+  // client cannot under-report what it actually occupied. When the
+  // server-side busy sampler is feeding add_busy(), the SAMPLED busy
+  // time is charged instead — a CPU-bound phase inside a lease then
+  // costs only the GPU time it actually used, and the client's report
+  // becomes advisory (quota-error measurement stays honest even
+  // against a hostile hook).
   void release(const std::string& pod, double used_ms, double now) {
     auto it = holders_.find(pod);
     if (it != holders_.end()) {
       double wall = now - it->second.grant_ms;
-      double charged = std::max(0.0, std::min(used_ms, wall + 50.0));
-      // a GPU-bound holder's wall time is the true exclusive occupancy
-      charged = std::max(charged, std::min(wall, it->second.quota));
+      double charged;
+      if (sampler_active_) {
+        charged = std::min(it->second.sampled_ms, wall + 50.0);
+      } else {
+        charged = std::max(0.0, std::min(used_ms, wall + 50.0));
+        // a GPU-bound holder's wall time is the true exclusive occupancy
+        charged = std::max(charged, std::min(wall, it->second.quota));
+      }
       account(pod).charge(now, charged);
       holders_.erase(it);
       if (holders_.empty()) holder_group_.clear();
@@ -136,6 +146,25 @@ class TokenScheduler {
     }
   }
 
+  // Server-side GPU-busy feed (gpu-schd samples rocm_smi's
+  // gpu_busy_percent between poll-loop ticks). Attributed to the
+  // current holder(s); co-granted gang members split the interval
+  // evenly (they share one GPU by construction). Unattributed busy
+  // (no holder: RCCL-exempt kernels, ungated processes) is tracked for
+  // stats only.
+  void add_busy(double busy_ms, double now) {
+    (void)now;
+    sampler_active_ = true;
+    if (holders_.empty()) {
+      other_busy_ms_ += busy_ms;
+      return;
+    }
+    double share = busy_ms / (double)holders_.size();
+    for (auto& kv : holders_) kv.second.sampled_ms += share;
+  }
+
+  bool sampler_active() const { return sampler_active_; }
+
   // Holder liveness: revoke tokens whose holders exceeded their lease
   // (died or hung; the connection may still be open).
   bool check_revoke(double now, double* revoked_at = nullptr) {
@@ -143,7 +172,9 @@ class TokenScheduler {
     for (auto it = holders_.begin(); it != holders_.end();) {
       double deadline = it->second.grant_ms + it->second.quota * 3.0 + 2000.0;
       if (now >= deadline) {
-        account(it->first).charge(now, now - it->second.grant_ms);
+        account(it->first).charge(
+            now, sampler_active_ ? it->second.sampled_ms
+                                 : now - it->second.grant_ms);
         it = holders_.erase(it);
         any = true;
       } else {
@@ -166,14 +197,41 @@ class TokenScheduler {
                    waiters_.end());
   }
 
-  // If a holder vanished without RET (connection closed), charge wall.
+  // If a holder vanished without RET (connection closed), charge wall
+  // (or sampled busy when the sampler runs).
   void force_release(const std::string& pod, double now) {
     auto it = holders_.find(pod);
     if (it != holders_.end()) {
-      account(pod).charge(now, now - it->second.grant_ms);
+      account(pod).charge(
+          now, sampler_active_ ? it->second.sampled_ms
+                               : now - it->second.grant_ms);
       holders_.erase(it);
       if (holders_.empty()) holder_group_.clear();
     }
+  }
+
+  // Record which connection's REQ produced a live grant, so a DYING
+  // connection only frees grants it owns. (Round-1 advisor finding:
+  // keying the cleanup by pod name could free a token a still-live
+  // connection of the same pod was holding — two concurrent holders.)
+  void set_holder_owner(const std::string& pod, long long owner) {
+    auto it = holders_.find(pod);
+    if (it != holders_.end()) it->second.owner = owner;
+  }
+
+  // A connection died: free exactly the grants it owns.
+  void force_release_owned(long long owner, double now) {
+    for (auto it = holders_.begin(); it != holders_.end();) {
+      if (it->second.owner == owner) {
+        account(it->first).charge(
+            now, sampler_active_ ? it->second.sampled_ms
+                                 : now - it->second.grant_ms);
+        it = holders_.erase(it);
+      } else {
+        ++it;
+      }
+    }
+    if (holders_.empty()) holder_group_.clear();
   }
 
   // Pick the next holder. Returns true and fills `out` when a token was
@@ -236,7 +294,15 @@ class TokenScheduler {
     PodAccount& a = account(w.pod);
     double room = a.limit >= 0.999 ? base_q_
                                    : a.limit * window_ - a.usage_cache;
-    double quota = std::min(base_q_, room);
+    // hint_ms = the client's EWMA of busy time per lease. A bursty pod
+    // that keeps releasing early gets right-sized (smaller) leases —
+    // finer-grained alternation, less over-grant — while the x1.5
+    // headroom lets a ramping pod grow back to base_q within a few
+    // leases. Always clamped to [min_q, base_q].
+    double desired = base_q_;
+    if (w.hint_ms > 0.0)
+      desired = std::min(base_q_, std::max(min_q_, w.hint_ms * 1.5));
+    double quota = std::min(desired, room);
     quota = std::max(quota, min_q_);
     holders_[w.pod] = Holder{now, quota};
     holder_group_ = a.group;
@@ -260,7 +326,10 @@ class TokenScheduler {
       total_busy += kv.second.usage_cache;
     }
     std::string s = "{\"window_ms\":" + fmt(window_) +
-                    ",\"busy_ms\":" + fmt(total_busy) + ",\"pods\":{";
+                    ",\"busy_ms\":" + fmt(total_busy) +
+                    ",\"sampler\":" + (sampler_active_ ? "true" : "false") +
+                    ",\"other_busy_ms\":" + fmt(other_busy_ms_) +
+                    ",\"pods\":{";
     bool first = true;
     for (auto& kv : pods_) {
       const PodAccount& a = kv.second;
@@ -290,6 +359,8 @@ class TokenScheduler {
   struct Holder {
     double grant_ms = 0.0;
     double quota = 0.0;
+    long long owner = -1;     // connection id whose REQ was granted
+    double sampled_ms = 0.0;  // server-sampled GPU-busy inside the lease
   };
 
   double base_q_, min_q_, window_;
@@ -297,6 +368,8 @@ class TokenScheduler {
   std::vector<Waiter> waiters_;
   std::map<std::string, Holder> holders_;
   std::string holder_group_;  // group of the current holders ("": none)
+  bool sampler_active_ = false;
+  double other_busy_ms_ = 0.0;  // sampled busy with no holder (stats)
 };
 
 }  // namespace ks

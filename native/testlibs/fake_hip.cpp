@@ -48,6 +48,29 @@ hipError_t hipMallocManaged(void** ptr, size_t size, unsigned int) {
   return hipMalloc(ptr, size);
 }
 
+// VMM family (PyTorch expandable_segments path) + pitched allocs — the
+// cap-bypass surfaces closed in round 2 (hiphook.cpp VMM section)
+hipError_t hipMemCreate(hipMemGenericAllocationHandle_t* handle, size_t size,
+                        const hipMemAllocationProp*, unsigned long long) {
+  *handle = (hipMemGenericAllocationHandle_t)malloc(16);
+  g_allocated += (long long)size;
+  return *handle ? hipSuccess : hipErrorOutOfMemory;
+}
+
+hipError_t hipMemRelease(hipMemGenericAllocationHandle_t handle) {
+  free((void*)handle);
+  return hipSuccess;
+}
+
+hipError_t hipMallocPitch(void** ptr, size_t* pitch, size_t width,
+                          size_t height) {
+  *pitch = (width + 255) / 256 * 256;  // model the HW pitch padding
+  size_t bytes = *pitch * height;
+  *ptr = malloc(bytes ? bytes : 1);
+  g_allocated += (long long)bytes;
+  return *ptr ? hipSuccess : hipErrorOutOfMemory;
+}
+
 hipError_t hipMemGetInfo(size_t* free_b, size_t* total_b) {
   if (total_b) *total_b = 16ull << 30;
   if (free_b) *free_b = (16ull << 30) - (size_t)g_allocated.load();

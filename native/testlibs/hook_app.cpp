@@ -85,9 +85,31 @@ static int scenario_rccl() {
   return 0;
 }
 
+static int scenario_vmm() {
+  // cap set by the test to 1 GiB; the VMM + pitched paths must honor it
+  hipMemGenericAllocationHandle_t h1 = nullptr, h2 = nullptr;
+  if (hipMemCreate(&h1, 512ull << 20, nullptr, 0) != hipSuccess) return 40;
+  if (hipMemCreate(&h2, 768ull << 20, nullptr, 0) != hipErrorOutOfMemory)
+    return 41;  // over cap: physical commit refused
+  // pitched alloc: 256 MiB worth of rows also counts against the cap
+  void* p = nullptr;
+  size_t pitch = 0;
+  if (hipMallocPitch(&p, &pitch, 1 << 20, 256) != hipSuccess) return 42;
+  if (pitch * 256 < (256ull << 20)) return 43;
+  // 512 + 256 = 768 MiB used; another 512 MiB VMM chunk must fail...
+  if (hipMemCreate(&h2, 512ull << 20, nullptr, 0) != hipErrorOutOfMemory)
+    return 44;
+  // ...until the first handle is released
+  if (hipMemRelease(h1) != hipSuccess) return 45;
+  if (hipMemCreate(&h2, 512ull << 20, nullptr, 0) != hipSuccess) return 46;
+  printf("VMM_OK\n");
+  return 0;
+}
+
 int main(int argc, char** argv) {
   if (argc < 2) return 2;
   if (strcmp(argv[1], "memcap") == 0) return scenario_memcap();
+  if (strcmp(argv[1], "vmm") == 0) return scenario_vmm();
   if (strcmp(argv[1], "gate") == 0)
     return scenario_gate(argc >= 3 ? atof(argv[2]) : 2000.0);
   if (strcmp(argv[1], "rccl") == 0) return scenario_rccl();
