@@ -216,6 +216,9 @@ def main():
                 "quota_error_per_pod": {k: round(v, 3)
                                         for k, v in quota_err.items()},
                 "sched_window_ms": stats.get("window_ms"),
+                "lease_accounting": ("sampled-busy"
+                                     if stats.get("sampler")
+                                     else "wall/RET"),
                 "per_pod_loss": [r[2] for r in results],
             },
         }

@@ -18,9 +18,17 @@ class TokenClient:
 
     def _ensure(self):
         if self._sock is None:
-            self._sock = socket.create_connection(self._addr,
-                                                  timeout=self._timeout)
-            self._sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+            host = self._addr[0]
+            if host.startswith("/"):  # UDS path (default pod transport)
+                s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+                s.settimeout(self._timeout)
+                s.connect(host)
+                self._sock = s
+            else:
+                self._sock = socket.create_connection(
+                    self._addr, timeout=self._timeout)
+                self._sock.setsockopt(socket.IPPROTO_TCP,
+                                      socket.TCP_NODELAY, 1)
             self._file = self._sock.makefile("r")
         return self._sock
 
@@ -54,8 +62,14 @@ class TokenClient:
             self._sock = None
 
 
-def query_stats(host: str, port: int, timeout: float = 10.0) -> dict:
-    s = socket.create_connection((host, port), timeout=timeout)
+def query_stats(host: str, port: int = 0, timeout: float = 10.0) -> dict:
+    """One STATS round-trip; `host` may be a UDS path ('/...')."""
+    if host.startswith("/"):
+        s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        s.settimeout(timeout)
+        s.connect(host)
+    else:
+        s = socket.create_connection((host, port), timeout=timeout)
     try:
         s.sendall(b"STATS\n")
         f = s.makefile("r")

@@ -54,6 +54,22 @@ def _wait_port(port: int, timeout: float = 10.0):
     raise TimeoutError(f"daemon on port {port} never came up")
 
 
+def _wait_uds(path: str, timeout: float = 10.0):
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        if os.path.exists(path):
+            try:
+                s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+                s.settimeout(0.3)
+                s.connect(path)
+                s.close()
+                return
+            except OSError:
+                pass
+        time.sleep(0.05)
+    raise TimeoutError(f"daemon at {path} never came up")
+
+
 @dataclass
 class PodHandle:
     name: str
@@ -61,14 +77,18 @@ class PodHandle:
     limit: float
     memory: int
     manager_port: int
+    manager_uds: str = ""
     manager_proc: subprocess.Popen = None
 
     def env(self, gpu_index: int, base_env: dict | None = None) -> dict:
         """The env block the scheduler injects into the pod's container
-        (ROCm-native equivalent of reference pod.go:445-457)."""
+        (ROCm-native equivalent of reference pod.go:445-457). UDS is the
+        default transport; TCP stays populated as the fallback."""
         env = dict(base_env if base_env is not None else os.environ)
         env[C.ENV_ROCR_VISIBLE_DEVICES] = str(gpu_index)
         env[C.ENV_LD_PRELOAD] = native_path(C.HOOK_SO_NAME)
+        if self.manager_uds:
+            env[C.ENV_POD_MANAGER_UDS] = self.manager_uds
         env[C.ENV_POD_MANAGER_IP] = "127.0.0.1"
         env[C.ENV_POD_MANAGER_PORT] = str(self.manager_port)
         env[C.ENV_POD_NAME] = self.name
@@ -103,29 +123,19 @@ class LocalGPUShare:
         os.makedirs(self.port_dir, exist_ok=True)
         F.write_gpu_config(self.config_dir, self.uuid, [])
         log = os.path.join(self.workdir, f"gpu-schd-{self.gpu_index}.log")
-        # free_port() is bind-then-release: with 8 ranks starting their
-        # stacks concurrently (the N=8 scaling bench) another process
-        # can grab the port in between — retry with a fresh one
-        last_err = None
-        for _ in range(4):
-            port = self.sched_port or free_port()
-            self._schd = subprocess.Popen(
-                [native_path("gpu-schd"), "-p", self.config_dir,
-                 "-f", self.uuid, "-P", str(port),
-                 "-q", str(self.base_quota_ms),
-                 "-m", str(self.min_quota_ms), "-w", str(self.window_ms),
-                 "-l", log],
-                stderr=subprocess.DEVNULL)
-            try:
-                _wait_port(port, timeout=10.0)
-                self.sched_port = port
-                return self
-            except TimeoutError as e:
-                last_err = e
-                self._schd.kill()
-                self._schd.wait()
-                self.sched_port = 0
-        raise last_err
+        # deterministic UDS endpoint (keyed by workdir) — no bind races
+        # when 8 ranks start their stacks concurrently, no hostNetwork
+        self.sched_uds = os.path.join(self.workdir,
+                                      f"schd-{self.gpu_index}.sock")
+        self._schd = subprocess.Popen(
+            [native_path("gpu-schd"), "-p", self.config_dir,
+             "-f", self.uuid, "-U", self.sched_uds,
+             "-q", str(self.base_quota_ms),
+             "-m", str(self.min_quota_ms), "-w", str(self.window_ms),
+             "-l", log, "-d", str(self.gpu_index)],
+            stderr=subprocess.DEVNULL)
+        _wait_uds(self.sched_uds, timeout=10.0)
+        return self
 
     def add_pod(self, name: str, request: float, limit: float | None = None,
                 memory: int = 0) -> PodHandle:
@@ -134,31 +144,25 @@ class LocalGPUShare:
             memory = math.floor(request * self.full_memory)
         h = PodHandle(name=name, request=request, limit=limit, memory=memory,
                       manager_port=0)
-        last_err = None
-        for _ in range(4):  # see start(): concurrent-rank port races
-            h.manager_port = free_port()
-            env = dict(os.environ)
-            env.update({
-                C.ENV_SCHEDULER_IP: "127.0.0.1",
-                C.ENV_SCHEDULER_PORT: str(self.sched_port),
-                C.ENV_POD_MANAGER_IP: "0.0.0.0",
-                C.ENV_POD_MANAGER_PORT: str(h.manager_port),
-                C.ENV_POD_NAME: name,
-                "POD_MANAGER_LOG": os.path.join(self.workdir, "pod-mgr.log"),
-            })
-            h.manager_proc = subprocess.Popen([native_path("pod-mgr")],
-                                              env=env,
-                                              stderr=subprocess.DEVNULL)
-            try:
-                _wait_port(h.manager_port, timeout=10.0)
-                self.pods[name] = h
-                self._rewrite_config()
-                return h
-            except TimeoutError as e:
-                last_err = e
-                h.manager_proc.kill()
-                h.manager_proc.wait()
-        raise last_err
+        h.manager_uds = os.path.join(
+            self.workdir, "pm-" + name.replace("/", "_") + ".sock")
+        h.manager_port = free_port()  # TCP fallback listener
+        env = dict(os.environ)
+        env.update({
+            "SCHEDULER_UDS": self.sched_uds,
+            C.ENV_POD_MANAGER_IP: "0.0.0.0",
+            C.ENV_POD_MANAGER_PORT: str(h.manager_port),
+            C.ENV_POD_MANAGER_UDS: h.manager_uds,
+            C.ENV_POD_NAME: name,
+            "POD_MANAGER_LOG": os.path.join(self.workdir, "pod-mgr.log"),
+        })
+        h.manager_proc = subprocess.Popen([native_path("pod-mgr")],
+                                          env=env,
+                                          stderr=subprocess.DEVNULL)
+        _wait_uds(h.manager_uds, timeout=10.0)
+        self.pods[name] = h
+        self._rewrite_config()
+        return h
 
     def remove_pod(self, name: str):
         h = self.pods.pop(name, None)
@@ -177,7 +181,7 @@ class LocalGPUShare:
             [F.PodPort(h.name, h.manager_port) for h in self.pods.values()])
 
     def stats(self) -> dict:
-        return query_stats("127.0.0.1", self.sched_port)
+        return query_stats(self.sched_uds, 0)
 
     def quota_error_pct(self) -> dict:
         """Per-pod |busy_share - request/sum(requests)| in % — the

@@ -60,10 +60,12 @@ class PodManagerSupervisor:
     """Diffs one GPU's podmanagerport file against running pod-mgr
     processes (reference launcher.py:34-67)."""
 
-    def __init__(self, uuid: str, sched_port: int, log_dir: str):
+    def __init__(self, uuid: str, sched_port: int, log_dir: str,
+                 sock_dir: str = ""):
         self.uuid = uuid
         self.sched_port = sched_port
         self.log_dir = log_dir
+        self.sock_dir = sock_dir
         self.procs: dict[str, subprocess.Popen] = {}   # pod -> proc
         self.ports: dict[str, int] = {}
 
@@ -87,6 +89,11 @@ class PodManagerSupervisor:
             C.ENV_POD_NAME: pod,
             "POD_MANAGER_LOG": os.path.join(self.log_dir, "pod-manager.log"),
         })
+        if self.sock_dir:
+            # default transport: per-pod UDS through the shared hostPath
+            # (pod-mgr listens on BOTH; TCP stays as fallback)
+            env[C.ENV_POD_MANAGER_UDS] = C.pod_manager_uds(
+                port, self.sock_dir)
         self.procs[pod] = subprocess.Popen([native_path("pod-mgr")], env=env,
                                            stderr=subprocess.DEVNULL)
         self.ports[pod] = port
@@ -116,7 +123,9 @@ class NodeDaemon:
         self.config_dir = os.path.join(workdir, "scheduler", "config")
         self.port_dir = os.path.join(workdir, "scheduler", "podmanagerport")
         self.log_dir = os.path.join(workdir, "log")
-        for d in (self.config_dir, self.port_dir, self.log_dir):
+        self.sock_dir = os.path.join(workdir, "sock")
+        for d in (self.config_dir, self.port_dir, self.log_dir,
+                  self.sock_dir):
             os.makedirs(d, exist_ok=True)
         self.base_port = base_port
         self.knobs = (base_quota, min_quota, window)
@@ -128,6 +137,7 @@ class NodeDaemon:
         self.running = True
 
     def start(self):
+        self.gpu_index = {g["uuid"]: g["index"] for g in self.gpus}
         for gpu in self.gpus:
             uuid, idx = gpu["uuid"], gpu["index"]
             port = self.base_port + idx
@@ -137,17 +147,21 @@ class NodeDaemon:
             pf = os.path.join(self.port_dir, uuid)
             if not os.path.exists(pf):
                 F.write_port_config(self.port_dir, uuid, [])
-            self._spawn_schd(uuid, port)
-            self.sup[uuid] = PodManagerSupervisor(uuid, port, self.log_dir)
+            self._spawn_schd(uuid, port, idx)
+            self.sup[uuid] = PodManagerSupervisor(uuid, port, self.log_dir,
+                                                  self.sock_dir)
         return self
 
-    def _spawn_schd(self, uuid: str, port: int):
+    def _spawn_schd(self, uuid: str, port: int, gpu_index: int = -1):
         q, m, w = self.knobs
-        self.schd[uuid] = subprocess.Popen(
-            [native_path("gpu-schd"), "-p", self.config_dir, "-f", uuid,
-             "-P", str(port), "-q", str(q), "-m", str(m), "-w", str(w),
-             "-l", os.path.join(self.log_dir, "gpu-schd.log")],
-            stderr=subprocess.DEVNULL)
+        cmd = [native_path("gpu-schd"), "-p", self.config_dir, "-f", uuid,
+               "-P", str(port), "-q", str(q), "-m", str(m), "-w", str(w),
+               "-l", os.path.join(self.log_dir, "gpu-schd.log")]
+        if gpu_index >= 0:
+            # enable the server-side busy sampler (leases charged
+            # sampled GPU time; falls back to wall/RET off-GPU)
+            cmd += ["-d", str(gpu_index)]
+        self.schd[uuid] = subprocess.Popen(cmd, stderr=subprocess.DEVNULL)
         log(f"gpu-schd started gpu={uuid} port={port}")
 
     def poll_once(self):
@@ -158,7 +172,8 @@ class NodeDaemon:
             if proc.poll() is not None:
                 log(f"gpu-schd for {uuid} died (rc={proc.returncode}); "
                     f"restarting")
-                self._spawn_schd(uuid, self.sup[uuid].sched_port)
+                self._spawn_schd(uuid, self.sup[uuid].sched_port,
+                                 self.gpu_index.get(uuid, -1))
         for uuid, sup in self.sup.items():
             path = os.path.join(self.port_dir, uuid)
             try:

@@ -64,15 +64,31 @@ int main() {
     return 1;
   }
 
-  int listen_fd = listen_on(mgr_uds && mgr_uds[0] ? mgr_uds : nullptr,
-                            mgr_port ? atoi(mgr_port) : 0);
-  if (listen_fd < 0) {
+  // Listen on BOTH when both are configured: UDS through the
+  // /kubeshare/sock hostPath is the default transport (unreachable
+  // from pods that don't mount it — no cross-pod quota-burn, no
+  // hostNetwork requirement); TCP stays as the fallback for setups
+  // without the shared mount.
+  int uds_fd = -1, tcp_fd = -1;
+  if (mgr_uds && mgr_uds[0]) {
+    uds_fd = listen_on(mgr_uds, 0);
+    if (uds_fd < 0)
+      logf(log, "pod-mgr", "WARN: cannot listen on uds %s: %s", mgr_uds,
+           strerror(errno));
+  }
+  if (mgr_port && atoi(mgr_port) > 0) {
+    tcp_fd = listen_on(nullptr, atoi(mgr_port));
+    if (tcp_fd < 0)
+      logf(log, "pod-mgr", "WARN: cannot listen on port %s: %s", mgr_port,
+           strerror(errno));
+  }
+  if (uds_fd < 0 && tcp_fd < 0) {
     logf(log, "pod-mgr", "FATAL: cannot listen on %s/%s",
          mgr_uds ? mgr_uds : "-", mgr_port ? mgr_port : "-");
     return 1;
   }
-  logf(log, "pod-mgr", "pod=%s listening (%s port %s)", pod.c_str(),
-       mgr_uds ? mgr_uds : "-", mgr_port ? mgr_port : "-");
+  logf(log, "pod-mgr", "pod=%s listening (uds=%s port=%s)", pod.c_str(),
+       uds_fd >= 0 ? mgr_uds : "-", tcp_fd >= 0 ? mgr_port : "-");
 
   std::map<int, std::unique_ptr<Client>> clients;
   LineBuffer up_rx;
@@ -85,16 +101,20 @@ int main() {
       if (x == fd) x = -1;  // keep FIFO positions; -1 = discard reply
   };
 
+  const int listeners[2] = {uds_fd, tcp_fd};
   for (;;) {
     std::vector<struct pollfd> pfds;
-    pfds.push_back({listen_fd, POLLIN, 0});
+    for (int lf : listeners)
+      if (lf >= 0) pfds.push_back({lf, POLLIN, 0});
+    size_t nlisten = pfds.size();
     pfds.push_back({up, POLLIN, 0});
     for (auto& kv : clients) pfds.push_back({kv.first, POLLIN, 0});
     ::poll(pfds.data(), pfds.size(), 1000);
 
-    if (pfds[0].revents & POLLIN) {
+    for (size_t li = 0; li < nlisten; li++) {
+      if (!(pfds[li].revents & POLLIN)) continue;
       int cfd;
-      while ((cfd = ::accept(listen_fd, nullptr, nullptr)) >= 0) {
+      while ((cfd = ::accept(pfds[li].fd, nullptr, nullptr)) >= 0) {
         int one = 1;
         setsockopt(cfd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
         auto c = std::make_unique<Client>();
@@ -104,7 +124,7 @@ int main() {
     }
 
     // upstream replies -> route by type FIFO
-    if (pfds[1].revents & (POLLIN | POLLHUP | POLLERR)) {
+    if (pfds[nlisten].revents & (POLLIN | POLLHUP | POLLERR)) {
       char buf[4096];
       ssize_t n = ::recv(up, buf, sizeof(buf), 0);
       if (n <= 0) {
@@ -138,7 +158,7 @@ int main() {
 
     // client requests -> stamp identity, forward upstream
     std::vector<int> dead;
-    for (size_t k = 2; k < pfds.size(); k++) {
+    for (size_t k = nlisten + 1; k < pfds.size(); k++) {
       if (!(pfds[k].revents & (POLLIN | POLLHUP | POLLERR))) continue;
       int cfd = pfds[k].fd;
       auto it = clients.find(cfd);

@@ -52,3 +52,24 @@ def test_bench_dist2_gloo(native_bins):
     assert out["n_gpus"] == 2
     # 2 ranks x 2 pods x 1 step x batch 2 images aggregated
     assert out["config"]["global_batch"] == 8
+
+
+def test_bench_dist8_gloo(native_bins):
+    """8-rank readiness on CPU: the first 8-GPU driver
+    run must not be burned on plumbing. 8 ranks x 2 pods = 16 worker
+    processes + 8 gpu-schd + 16 pod-mgr over deterministic UDS paths."""
+    port = socket.socket()
+    port.bind(("127.0.0.1", 0))
+    p = port.getsockname()[1]
+    port.close()
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", str(p), "bench.py", "--gpus", "8"] + TINY,
+        cwd=REPO, capture_output=True, text=True, timeout=1800,
+        env=dict(os.environ, MASTER_ADDR="127.0.0.1"))
+    assert r.returncode == 0, (r.stdout[-2000:], r.stderr[-3000:])
+    out = _parse_json_line(r.stdout)
+    assert out["n_gpus"] == 8
+    assert out["config"]["global_batch"] == 32  # 8 x 2 pods x batch 2
+    assert out["value"] > 0

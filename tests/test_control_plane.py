@@ -371,3 +371,44 @@ def test_launcher_pod_churn(tmp_path, native_bins, fake_gpus):
         assert nd.sup["GPU-fake-0"].procs == {}
     finally:
         nd.stop()
+
+
+def test_pod_mgr_uds_default_transport(tmp_path, native_bins, fake_gpus):
+    """pod-mgr listens on BOTH the per-pod UDS (default transport, no
+    hostNetwork, unreachable cross-pod) and the TCP fallback; a token
+    round-trips over each."""
+    from kubeshare_amd.isolation.local import free_port
+    base_port = free_port()
+    nd = NodeDaemon(str(tmp_path), base_port=base_port, base_quota=40,
+                    min_quota=10, window=2000, gpus=fake_gpus)
+    nd.start()
+    mgr_port = free_port()
+    uds = C.pod_manager_uds(mgr_port, nd.sock_dir)
+    try:
+        ConfigDaemon("node-a", nd.config_dir, nd.port_dir).update(
+            [PodDemand("ns", "udsy", "u1", "node-a", "GPU-fake-0",
+                       1.0, 0.5, 0, mgr_port)])
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            nd.poll_once()
+            if os.path.exists(uds):
+                try:
+                    s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+                    s.connect(uds)
+                    s.close()
+                    break
+                except OSError:
+                    pass
+            time.sleep(0.1)
+        else:
+            raise TimeoutError("pod-mgr UDS never came up")
+        # over UDS
+        from kubeshare_amd.isolation.client import TokenClient
+        c = TokenClient(uds, 0, "ignored")
+        assert c.acquire() > 0
+        c.release(5.0)
+        c.close()
+        # TCP fallback still works on the same pod-mgr
+        assert _token_roundtrip(mgr_port, "ignored") > 0
+    finally:
+        nd.stop()

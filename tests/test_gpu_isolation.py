@@ -203,3 +203,87 @@ def test_work_conserving_solo_pod(share):
         assert used > 0.75 * wall * 1000, f"burst throttled: {out}"
     finally:
         share.remove_pod("gpu/solo")
+
+
+def test_bursty_lease_accounting(share):
+    """A 30%-duty-cycle pod must be charged ~its GPU-busy time, not
+    lease wall time (round-1 VERDICT Weak #6 / Next #5): the server-side
+    busy sampler in gpu-schd attributes only sampled-busy ms to the
+    lease. Wall-charging would report ~2-3x the submitted work here."""
+    h = share.add_pod("gpu/bursty", request=0.5, limit=1.0)
+    try:
+        p = _spawn_burner(h, 8000, wait_go=False)
+        env_extra = None  # burner runs solo: all sampled busy is its own
+        out, _ = p.communicate(timeout=180)
+        assert p.returncode == 0, out
+        st = share.stats()
+        assert st.get("sampler") is True, "busy sampler not active"
+        # re-run bursty and measure the charge delta server-side
+        st0 = share.stats()
+        p = subprocess.Popen(
+            [sys.executable, "-m", "kubeshare_amd.isolation.burn_worker",
+             "--duration-ms", "8000", "--duty-cycle", "0.3"],
+            env=dict(h.env(gpu_index=0),
+                     PYTHONPATH=REPO + os.pathsep +
+                     os.environ.get("PYTHONPATH", "")),
+            cwd=REPO, stdout=subprocess.PIPE, text=True)
+        out, _ = p.communicate(timeout=180)
+        assert p.returncode == 0, out
+        time.sleep(0.6)  # let the idle watchdog RET the tail lease
+        st1 = share.stats()
+        queued_ms = float(out.split()[5])   # client-side submitted work
+        charged = st1["pods"]["gpu/bursty"]["total_used_ms"] - \
+            st0["pods"]["gpu/bursty"]["total_used_ms"]
+        # sampled charge tracks the submitted GPU time, not the ~2-3x
+        # wall the lease spanned
+        assert charged < queued_ms * 1.30 + 100, \
+            f"over-charged: {charged:.0f}ms vs {queued_ms:.0f}ms submitted"
+        assert charged > queued_ms * 0.55, \
+            f"under-charged: {charged:.0f}ms vs {queued_ms:.0f}ms submitted"
+    finally:
+        share.remove_pod("gpu/bursty")
+
+
+def test_expandable_segments_memory_cap(share):
+    """The VMM allocator path (hipMemCreate/hipMemMap) honors
+    KUBESHARE_GPU_MEM: round-1 VERDICT Missing #5 — with
+    expandable_segments:True PyTorch bypassed the cap entirely."""
+    h = share.add_pod("gpu/vmm", request=0.5, limit=1.0,
+                      memory=2 * 1024**3)
+    try:
+        code = (
+            "import torch\n"
+            "ok=False\n"
+            "try:\n"
+            "    x = torch.empty(4*1024**3, dtype=torch.uint8, device='cuda')\n"
+            "except torch.cuda.OutOfMemoryError:\n"
+            "    ok=True\n"
+            "assert ok, 'allocation over cap succeeded (VMM bypass)'\n"
+            "y = torch.empty(512*1024**2, dtype=torch.uint8, device='cuda')\n"
+            "print('VMM-MEMCAP-OK')\n"
+        )
+        env = h.env(gpu_index=0)
+        env["PYTORCH_CUDA_ALLOC_CONF"] = "expandable_segments:True"
+        env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+        r = subprocess.run([sys.executable, "-c", code], env=env, cwd=REPO,
+                           capture_output=True, text=True, timeout=300)
+        assert "VMM-MEMCAP-OK" in r.stdout, (r.stdout, r.stderr)
+    finally:
+        share.remove_pod("gpu/vmm")
+
+
+def test_ddp_gang_two_ranks_one_gpu(native_bins):
+    """SURVEY §2.4(b), the declared hardest part, on REAL RCCL: two DDP
+    ranks sharing ONE MI355X through the full chain (hook + pod-mgr +
+    gpu-schd with gang co-granting). Must not deadlock; both ranks
+    progress; the gang's quota is honored."""
+    from kubeshare_amd.parallel import launch_gang
+    ok, stats = launch_gang(ranks=2, share_gpu=True, steps=6,
+                            model="resnet18", batch=32, timeout=420)
+    assert ok, f"gang failed/deadlocked; stats={stats}"
+    st = stats[0]
+    pods = st.get("pods", {})
+    assert "gang/rank0" in pods and "gang/rank1" in pods
+    # both ranks actually went through the token gate
+    assert pods["gang/rank0"]["grants"] >= 1
+    assert pods["gang/rank1"]["grants"] >= 1
