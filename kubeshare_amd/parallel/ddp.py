@@ -47,7 +47,14 @@ def ddp_worker(model_name: str = "resnet18", batch: int = 32,
     from ..models import build_model
 
     on_gpu = torch.cuda.is_available()
-    dist.init_process_group("nccl" if on_gpu else "gloo")
+    # RCCL refuses two ranks on one device ("Duplicate GPU detected" —
+    # same restriction as NCCL; the reference's own gang workloads only
+    # ever put one NCCL rank per GPU, test/distribute/*). For the
+    # shared-GPU gang config the launcher therefore selects gloo
+    # gradients over GPU compute; whole-GPU gangs use RCCL.
+    backend = os.environ.get("KUBESHARE_DDP_BACKEND") or \
+        ("nccl" if on_gpu else "gloo")
+    dist.init_process_group(backend)
     dev = "cuda" if on_gpu else "cpu"
     if on_gpu:
         torch.cuda.set_device(0)  # ROCR_VISIBLE_DEVICES narrows the view
@@ -120,6 +127,8 @@ def launch_gang(ranks: int = 2, share_gpu: bool = False, steps: int = 10,
                 "PYTHONPATH": _REPO + os.pathsep +
                 env.get("PYTHONPATH", ""),
             })
+            if share_gpu:
+                env["KUBESHARE_DDP_BACKEND"] = "gloo"  # see ddp_worker
             procs.append(subprocess.Popen(
                 [sys.executable, "-c",
                  "from kubeshare_amd.parallel.ddp import ddp_worker; "

@@ -273,10 +273,13 @@ def test_expandable_segments_memory_cap(share):
 
 
 def test_ddp_gang_two_ranks_one_gpu(native_bins):
-    """SURVEY §2.4(b), the declared hardest part, on REAL RCCL: two DDP
-    ranks sharing ONE MI355X through the full chain (hook + pod-mgr +
-    gpu-schd with gang co-granting). Must not deadlock; both ranks
-    progress; the gang's quota is honored."""
+    """SURVEY §2.4(b) on hardware: two token-gated DDP ranks sharing
+    ONE MI355X through the full chain (hook + pod-mgr + gpu-schd with
+    gang co-granting). Gradient collectives use gloo — RCCL, like NCCL,
+    refuses two ranks on one device ("Duplicate GPU detected"; the
+    reference's gang workloads likewise run one NCCL rank per GPU) —
+    the GPU compute is fully gated either way. Must not deadlock; both
+    ranks progress; both pass the token gate."""
     from kubeshare_amd.parallel import launch_gang
     ok, stats = launch_gang(ranks=2, share_gpu=True, steps=6,
                             model="resnet18", batch=32, timeout=420)
@@ -287,3 +290,44 @@ def test_ddp_gang_two_ranks_one_gpu(native_bins):
     # both ranks actually went through the token gate
     assert pods["gang/rank0"]["grants"] >= 1
     assert pods["gang/rank1"]["grants"] >= 1
+
+
+def test_rccl_communicator_under_sharing(share):
+    """REAL librccl under the token gate: a rank holding an RCCL
+    communicator runs gated compute + ungated collectives while a
+    co-located burner pod competes for the GPU. The librccl call-site
+    exemption (hiphook.cpp exempt_caller) must keep collectives off the
+    gate — a gated collective here would stall behind the burner's
+    token and eventually deadlock a real multi-GPU gang."""
+    h = share.add_pod("gpu/rcclrank", request=0.5, limit=1.0)
+    comp = share.add_pod("gpu/competitor", request=0.5, limit=1.0)
+    code = (
+        "import os, torch, torch.distributed as dist, ctypes\n"
+        "dist.init_process_group('nccl', rank=0, world_size=1)\n"
+        "x = torch.ones(1 << 20, device='cuda')\n"
+        "for i in range(40):\n"
+        "    x = x * 1.0000001\n"
+        "    dist.all_reduce(x)\n"
+        "torch.cuda.synchronize()\n"
+        "assert torch.isfinite(x).all()\n"
+        "lib = ctypes.CDLL(None)\n"
+        "lib.ks_hook_leases.restype = ctypes.c_longlong\n"
+        "assert lib.ks_hook_leases() >= 1, 'gate never engaged'\n"
+        "print('RCCL-SHARING-OK', lib.ks_hook_leases())\n"
+        "dist.destroy_process_group()\n"
+    )
+    try:
+        burner = _spawn_burner(comp, 20000)
+        env = h.env(gpu_index=0)
+        env.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29581",
+                    "RANK": "0", "WORLD_SIZE": "1",
+                    "PYTHONPATH": REPO + os.pathsep +
+                    env.get("PYTHONPATH", "")})
+        r = subprocess.run([sys.executable, "-c", code], env=env, cwd=REPO,
+                           capture_output=True, text=True, timeout=300)
+        assert "RCCL-SHARING-OK" in r.stdout, (r.stdout[-2000:],
+                                               r.stderr[-2000:])
+        burner.communicate(timeout=120)
+    finally:
+        share.remove_pod("gpu/rcclrank")
+        share.remove_pod("gpu/competitor")
