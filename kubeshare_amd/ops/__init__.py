@@ -135,13 +135,13 @@ def _bn_relu_autograd():
                 y, mean, invstd = ext.bn_relu_fwd_train(
                     x, w32, b32, running_mean, running_var, momentum, eps,
                     res)
-                ctx.save_for_backward(x, y, w32, mean, invstd)
+                ctx.save_for_backward(x, y, w32, b32, mean, invstd)
                 ctx.with_res = res is not None
                 ctx.wdtype = weight.dtype
                 return y
             y = ext.bn_relu_fwd_eval(x, w32, b32, running_mean.float(),
                                      running_var.float(), eps, res)
-            ctx.save_for_backward(x, y, w32, running_mean.float(),
+            ctx.save_for_backward(x, y, w32, b32, running_mean.float(),
                                   (running_var.float() + eps).rsqrt())
             ctx.with_res = res is not None
             ctx.wdtype = weight.dtype
@@ -150,8 +150,9 @@ def _bn_relu_autograd():
         @staticmethod
         def backward(ctx, dy):
             ext = _load()
-            x, y, w32, mean, invstd = ctx.saved_tensors
-            out = ext.bn_relu_bwd(x, y, dy, w32, mean, invstd, ctx.with_res)
+            x, y, w32, b32, mean, invstd = ctx.saved_tensors
+            out = ext.bn_relu_bwd(x, y, dy, w32, b32, mean, invstd,
+                                  ctx.with_res)
             dx, dscale, dbias = out[0], out[1], out[2]
             dres = out[3] if ctx.with_res else None
             return (dx, dres, dscale.to(ctx.wdtype), dbias.to(ctx.wdtype),

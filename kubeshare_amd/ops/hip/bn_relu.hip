@@ -340,6 +340,239 @@ __global__ void bn_bwd_stats_kernel(const ushort8* __restrict__ x,
   }
 }
 
+// ----------------------------------------- bwd stats, no-y variant
+// Non-residual BNs: the ReLU mask is recomputed from x and the folded
+// per-channel constants (y>0 <=> bf16(relu(s*x+b)) != 0, replicating
+// the forward's rounding exactly), so the backward never re-reads y:
+// 7 tensor passes -> 5 for the non-residual blocks of ResNet50.
+__global__ void bn_bwd_stats_noy_kernel(const ushort8* __restrict__ x,
+                                        const ushort8* __restrict__ dy,
+                                        const float* __restrict__ mean,
+                                        const float* __restrict__ invstd,
+                                        const float* __restrict__ weight,
+                                        const float* __restrict__ bias,
+                                        float* __restrict__ part,
+                                        long long M, int CG) {
+  __shared__ float s_red[KS_BN_BLOCK * 8];
+  const int tid = threadIdx.x;
+  const int cg = tid % CG;
+  const int roff = tid / CG;
+  const int rows_per_blk = KS_BN_BLOCK / CG;
+  const bool active = roff < rows_per_blk;  // see bn_stats_kernel
+  const long long stride = (long long)gridDim.x * rows_per_blk;
+  float8 mu, is, sc, bf;
+#pragma unroll
+  for (int j = 0; j < 8; j++) {
+    const int c = cg * 8 + j;
+    mu[j] = mean[c];
+    is[j] = invstd[c];
+    sc[j] = weight[c] * is[j];           // folded scale (as in fwd)
+    bf[j] = bias[c] - mu[j] * sc[j];     // folded bias
+  }
+  float8 adb = {0, 0, 0, 0, 0, 0, 0, 0};
+  float8 ads = {0, 0, 0, 0, 0, 0, 0, 0};
+  long long row = active ? (long long)blockIdx.x * rows_per_blk + roff : M;
+  for (; row + (KS_BN_UNROLL - 1) * stride < M;
+       row += KS_BN_UNROLL * stride) {
+    ushort8 xv[KS_BN_UNROLL], gv[KS_BN_UNROLL];
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++) {
+      const long long k = (row + u * stride) * CG + cg;
+      xv[u] = x[k];
+      gv[u] = dy[k];
+    }
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++)
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float xf = bf16_to_f32(xv[u][j]);
+        float f = fmaf(xf, sc[j], bf[j]);
+        // exact fwd replication: y stored as bf16(relu(f))
+        float g = f32_to_bf16(f > 0.f ? f : 0.f)
+                      ? bf16_to_f32(gv[u][j]) : 0.f;
+        adb[j] += g;
+        ads[j] += g * ((xf - mu[j]) * is[j]);
+      }
+  }
+  for (; row < M; row += stride) {
+    const long long k = row * CG + cg;
+    ushort8 xv = x[k], gv = dy[k];
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float xf = bf16_to_f32(xv[j]);
+      float f = fmaf(xf, sc[j], bf[j]);
+      float g = f32_to_bf16(f > 0.f ? f : 0.f) ? bf16_to_f32(gv[j]) : 0.f;
+      adb[j] += g;
+      ads[j] += g * ((xf - mu[j]) * is[j]);
+    }
+  }
+
+  const int C = CG * 8;
+  float* db_part = part + (long long)blockIdx.x * C;
+  float* ds_part = part + (long long)(gridDim.x + blockIdx.x) * C;
+#pragma unroll
+  for (int j = 0; j < 8; j++) s_red[tid * 8 + j] = adb[j];
+  __syncthreads();
+  if (tid < CG) {
+    float8 t = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int r = 0; r < rows_per_blk; r++)
+#pragma unroll
+      for (int j = 0; j < 8; j++) t[j] += s_red[(r * CG + tid) * 8 + j];
+#pragma unroll
+    for (int j = 0; j < 8; j++) db_part[tid * 8 + j] = t[j];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int j = 0; j < 8; j++) s_red[tid * 8 + j] = ads[j];
+  __syncthreads();
+  if (tid < CG) {
+    float8 t = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int r = 0; r < rows_per_blk; r++)
+#pragma unroll
+      for (int j = 0; j < 8; j++) t[j] += s_red[(r * CG + tid) * 8 + j];
+#pragma unroll
+    for (int j = 0; j < 8; j++) ds_part[tid * 8 + j] = t[j];
+  }
+}
+
+// ----------------------------------------- bwd apply, no-y variant
+__global__ void bn_bwd_apply_noy_kernel(const ushort8* __restrict__ x,
+                                        const ushort8* __restrict__ dy,
+                                        ushort8* __restrict__ dx,
+                                        const float* __restrict__ mean,
+                                        const float* __restrict__ invstd,
+                                        const float* __restrict__ weight,
+                                        const float* __restrict__ bias,
+                                        const float* __restrict__ dbias,
+                                        const float* __restrict__ dscale,
+                                        long long M, int CG) {
+  const int tid = threadIdx.x;
+  const int cg = tid % CG;
+  const int roff = tid / CG;
+  const int rows_per_blk = KS_BN_BLOCK / CG;
+  const bool active = roff < rows_per_blk;
+  const long long stride = (long long)gridDim.x * rows_per_blk;
+  const float invM = 1.f / (float)M;
+  float8 mu, is, sc, bf, w, db, ds;
+#pragma unroll
+  for (int j = 0; j < 8; j++) {
+    const int c = cg * 8 + j;
+    mu[j] = mean[c];
+    is[j] = invstd[c];
+    sc[j] = weight[c] * is[j];
+    bf[j] = bias[c] - mu[j] * sc[j];
+    w[j] = sc[j];
+    db[j] = dbias[c] * invM;
+    ds[j] = dscale[c] * invM;
+  }
+  long long row = active ? (long long)blockIdx.x * rows_per_blk + roff : M;
+  for (; row + (KS_BN_UNROLL - 1) * stride < M;
+       row += KS_BN_UNROLL * stride) {
+    ushort8 xv[KS_BN_UNROLL], gv[KS_BN_UNROLL];
+    long long k[KS_BN_UNROLL];
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++) {
+      k[u] = (row + u * stride) * CG + cg;
+      xv[u] = x[k[u]];
+      gv[u] = dy[k[u]];
+    }
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++) {
+      ushort8 o;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float xf = bf16_to_f32(xv[u][j]);
+        float f = fmaf(xf, sc[j], bf[j]);
+        float g = f32_to_bf16(f > 0.f ? f : 0.f)
+                      ? bf16_to_f32(gv[u][j]) : 0.f;
+        float xhat = (xf - mu[j]) * is[j];
+        o[j] = f32_to_bf16(w[j] * (g - db[j] - xhat * ds[j]));
+      }
+      dx[k[u]] = o;
+    }
+  }
+  for (; row < M; row += stride) {
+    const long long k = row * CG + cg;
+    ushort8 xv = x[k], gv = dy[k];
+    ushort8 o;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float xf = bf16_to_f32(xv[j]);
+      float f = fmaf(xf, sc[j], bf[j]);
+      float g = f32_to_bf16(f > 0.f ? f : 0.f) ? bf16_to_f32(gv[j]) : 0.f;
+      float xhat = (xf - mu[j]) * is[j];
+      o[j] = f32_to_bf16(w[j] * (g - db[j] - xhat * ds[j]));
+    }
+    dx[k] = o;
+  }
+}
+
+// ----------------------------------------- bwd apply from dym
+// Residual BNs: the stats pass already wrote dym (= dres); read it
+// instead of y AND dy — 8 tensor passes -> 7.
+__global__ void bn_bwd_apply_dym_kernel(const ushort8* __restrict__ x,
+                                        const ushort8* __restrict__ dym,
+                                        ushort8* __restrict__ dx,
+                                        const float* __restrict__ mean,
+                                        const float* __restrict__ invstd,
+                                        const float* __restrict__ weight,
+                                        const float* __restrict__ dbias,
+                                        const float* __restrict__ dscale,
+                                        long long M, int CG) {
+  const int tid = threadIdx.x;
+  const int cg = tid % CG;
+  const int roff = tid / CG;
+  const int rows_per_blk = KS_BN_BLOCK / CG;
+  const bool active = roff < rows_per_blk;
+  const long long stride = (long long)gridDim.x * rows_per_blk;
+  const float invM = 1.f / (float)M;
+  float8 mu, is, w, db, ds;
+#pragma unroll
+  for (int j = 0; j < 8; j++) {
+    const int c = cg * 8 + j;
+    mu[j] = mean[c];
+    is[j] = invstd[c];
+    w[j] = weight[c] * is[j];
+    db[j] = dbias[c] * invM;
+    ds[j] = dscale[c] * invM;
+  }
+  long long row = active ? (long long)blockIdx.x * rows_per_blk + roff : M;
+  for (; row + (KS_BN_UNROLL - 1) * stride < M;
+       row += KS_BN_UNROLL * stride) {
+    ushort8 xv[KS_BN_UNROLL], gv[KS_BN_UNROLL];
+    long long k[KS_BN_UNROLL];
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++) {
+      k[u] = (row + u * stride) * CG + cg;
+      xv[u] = x[k[u]];
+      gv[u] = dym[k[u]];
+    }
+#pragma unroll
+    for (int u = 0; u < KS_BN_UNROLL; u++) {
+      ushort8 o;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float g = bf16_to_f32(gv[u][j]);  // already ReLU-masked
+        float xhat = (bf16_to_f32(xv[u][j]) - mu[j]) * is[j];
+        o[j] = f32_to_bf16(w[j] * (g - db[j] - xhat * ds[j]));
+      }
+      dx[k[u]] = o;
+    }
+  }
+  for (; row < M; row += stride) {
+    const long long k = row * CG + cg;
+    ushort8 xv = x[k], gv = dym[k];
+    ushort8 o;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float g = bf16_to_f32(gv[j]);
+      float xhat = (bf16_to_f32(xv[j]) - mu[j]) * is[j];
+      o[j] = f32_to_bf16(w[j] * (g - db[j] - xhat * ds[j]));
+    }
+    dx[k] = o;
+  }
+}
+
 // --------------------------------------------------------- bwd apply
 // dx = w*invstd * (dym - dbias/M - xhat * dscale/M)
 __global__ void bn_bwd_apply_kernel(const ushort8* __restrict__ x,
@@ -544,7 +777,7 @@ torch::Tensor bn_relu_fwd_eval(torch::Tensor x, torch::Tensor weight,
 
 std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor x, torch::Tensor y,
                                        torch::Tensor dy, torch::Tensor weight,
-                                       torch::Tensor mean,
+                                       torch::Tensor bias, torch::Tensor mean,
                                        torch::Tensor invstd, bool need_dres) {
   Geom g = geom_of(x);
   auto stream = at::cuda::getCurrentCUDAStream();
@@ -559,6 +792,8 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor x, torch::Tensor y,
     dy = dy.contiguous(at::MemoryFormat::ChannelsLast);
 
   if (need_dres) {
+    // residual: the stats pass writes dym (= dres), the apply pass
+    // reads it back instead of y AND dy (8 tensor passes -> 7)
     dres = at::empty_like(x);
     hipLaunchKernelGGL(bn_bwd_stats_kernel<true>, dim3(g.stat_blocks),
                        dim3(KS_BN_BLOCK), 0, stream.stream(),
@@ -568,31 +803,44 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor x, torch::Tensor y,
                        reinterpret_cast<ushort8*>(dres.data_ptr()),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        part.data_ptr<float>(), g.M, g.CG);
-  } else {
-    hipLaunchKernelGGL(bn_bwd_stats_kernel<false>, dim3(g.stat_blocks),
+    hipLaunchKernelGGL(bn_reduce_partials_kernel,
+                       dim3((g.C + 3) / 4), dim3(64, 4), 0, stream.stream(),
+                       part.data_ptr<float>(), g.stat_blocks, g.C,
+                       dbias_p, dscale_p);
+    hipLaunchKernelGGL(bn_bwd_apply_dym_kernel, dim3(g.blocks),
                        dim3(KS_BN_BLOCK), 0, stream.stream(),
                        reinterpret_cast<const ushort8*>(x.data_ptr()),
-                       reinterpret_cast<const ushort8*>(y.data_ptr()),
-                       reinterpret_cast<const ushort8*>(dy.data_ptr()),
-                       nullptr, mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), part.data_ptr<float>(),
+                       reinterpret_cast<const ushort8*>(dres.data_ptr()),
+                       reinterpret_cast<ushort8*>(dx.data_ptr()),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       weight.data_ptr<float>(), dbias_p, dscale_p,
                        g.M, g.CG);
+    auto dbias = red.narrow(0, 0, g.C);
+    auto dscale = red.narrow(0, g.C, g.C);
+    return {dx, dscale, dbias, dres};
   }
+  // non-residual: the ReLU mask is recomputed from x and the folded
+  // scale/bias, so y is never read (7 tensor passes -> 5)
+  hipLaunchKernelGGL(bn_bwd_stats_noy_kernel, dim3(g.stat_blocks),
+                     dim3(KS_BN_BLOCK), 0, stream.stream(),
+                     reinterpret_cast<const ushort8*>(x.data_ptr()),
+                     reinterpret_cast<const ushort8*>(dy.data_ptr()),
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     weight.data_ptr<float>(), bias.data_ptr<float>(),
+                     part.data_ptr<float>(), g.M, g.CG);
   hipLaunchKernelGGL(bn_reduce_partials_kernel,
                      dim3((g.C + 3) / 4), dim3(64, 4), 0, stream.stream(),
                      part.data_ptr<float>(), g.stat_blocks, g.C,
                      dbias_p, dscale_p);
-  hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(g.blocks), dim3(KS_BN_BLOCK),
-                     0, stream.stream(),
+  hipLaunchKernelGGL(bn_bwd_apply_noy_kernel, dim3(g.blocks),
+                     dim3(KS_BN_BLOCK), 0, stream.stream(),
                      reinterpret_cast<const ushort8*>(x.data_ptr()),
-                     reinterpret_cast<const ushort8*>(y.data_ptr()),
                      reinterpret_cast<const ushort8*>(dy.data_ptr()),
                      reinterpret_cast<ushort8*>(dx.data_ptr()),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                     weight.data_ptr<float>(), dbias_p, dscale_p, g.M, g.CG);
+                     weight.data_ptr<float>(), bias.data_ptr<float>(),
+                     dbias_p, dscale_p, g.M, g.CG);
   auto dbias = red.narrow(0, 0, g.C);
   auto dscale = red.narrow(0, g.C, g.C);
-  if (need_dres)
-    return {dx, dscale, dbias, dres};
   return {dx, dscale, dbias};
 }

@@ -129,7 +129,7 @@ torch::Tensor bn_relu_fwd_eval(torch::Tensor x, torch::Tensor weight,
                                c10::optional<torch::Tensor> res);
 std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor x, torch::Tensor y,
                                        torch::Tensor dy, torch::Tensor weight,
-                                       torch::Tensor mean,
+                                       torch::Tensor bias, torch::Tensor mean,
                                        torch::Tensor invstd, bool need_dres);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
