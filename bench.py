@@ -219,6 +219,8 @@ def main():
                 "lease_accounting": ("sampled-busy"
                                      if stats.get("sampler")
                                      else "wall/RET"),
+                "grants_per_pod": {p: v.get("grants")
+                                   for p, v in stats["pods"].items()},
                 "per_pod_loss": [r[2] for r in results],
             },
         }

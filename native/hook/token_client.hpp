@@ -138,6 +138,8 @@ class TokenGate {
     drain_arg_ = drain_arg;
     if (const char* e = getenv("KUBESHARE_IDLE_RELEASE_MS"))
       idle_release_ms_ = atof(e);
+    if (const char* e = getenv("KUBESHARE_LEASE_HINT"))
+      hint_enabled_ = e[0] != '0';
     enabled_.store(client_.configured(), std::memory_order_release);
   }
 
@@ -271,8 +273,8 @@ class TokenGate {
     start_watchdog_locked();
     // hint = EWMA of recent per-lease use, so gpu-schd can right-size
     // the next quota (a bursty pod gets shorter leases; see
-    // token_sched.hpp schedule())
-    double quota = client_.acquire(ewma_used_);
+    // token_sched.hpp schedule()). KUBESHARE_LEASE_HINT=0 disables.
+    double quota = client_.acquire(hint_enabled_ ? ewma_used_ : 0.0);
     if (quota <= 0.0) {
       // Scheduler unreachable: fail OPEN for liveness (the node daemon
       // restarts the chain; isolation degrades, jobs don't die) unless
@@ -300,6 +302,7 @@ class TokenGate {
   std::atomic<double> last_activity_{0.0};
   double idle_release_ms_ = 25.0;
   bool holding_ = false;
+  bool hint_enabled_ = true;
   double grant_time_ = 0.0;
   double ewma_used_ = 0.0;
 
