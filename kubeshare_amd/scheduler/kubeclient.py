@@ -214,12 +214,44 @@ class RestCoreV1:
 
     def patch_namespaced_pod(self, name: str, namespace: str, body: dict):
         # strategic-merge patch is enough for annotation updates
-        url = f"/api/v1/namespaces/{namespace}/pods/{name}"
+        return self._merge_patch(
+            f"/api/v1/namespaces/{namespace}/pods/{name}", body)
+
+    def _merge_patch(self, path: str, body: dict):
         data = json.dumps(body).encode()
-        req = urllib.request.Request(self.base_url + url, data=data,
+        req = urllib.request.Request(self.base_url + path, data=data,
                                      method="PATCH")
         req.add_header("Content-Type", "application/merge-patch+json")
         if self.token:
             req.add_header("Authorization", f"Bearer {self.token}")
         with urllib.request.urlopen(req, context=self._ctx, timeout=30) as r:
             return json.loads(r.read() or b"{}")
+
+
+class RestCustomObjects:
+    """CustomObjectsApi-compatible subset (SharePod controller)."""
+
+    def __init__(self, core: RestCoreV1):
+        self.core = core
+
+    def list_cluster_custom_object(self, group, version, plural):
+        return self.core._req("GET", f"/apis/{group}/{version}/{plural}")
+
+    def create_namespaced_custom_object(self, group, version, namespace,
+                                        plural, body):
+        return self.core._req(
+            "POST", f"/apis/{group}/{version}/namespaces/{namespace}/"
+            f"{plural}", body=body)
+
+    def get_namespaced_custom_object(self, group, version, namespace,
+                                     plural, name):
+        return self.core._req(
+            "GET", f"/apis/{group}/{version}/namespaces/{namespace}/"
+            f"{plural}/{name}")
+
+    def patch_namespaced_custom_object_status(self, group, version,
+                                              namespace, plural, name,
+                                              body):
+        return self.core._merge_patch(
+            f"/apis/{group}/{version}/namespaces/{namespace}/{plural}/"
+            f"{name}/status", body)

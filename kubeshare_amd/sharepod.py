@@ -59,14 +59,22 @@ class SharePodController:
 
     GROUP, VERSION, PLURAL = "sharedgpu.kubeshare.amd", "v1", "sharepods"
 
-    def __init__(self):
-        from kubernetes import client, config
+    def __init__(self, v1=None, crd=None):
+        if v1 is not None and crd is not None:
+            self.v1, self.crd = v1, crd
+            return
         try:
-            config.load_incluster_config()
-        except Exception:  # noqa: BLE001
-            config.load_kube_config()
-        self.v1 = client.CoreV1Api()
-        self.crd = client.CustomObjectsApi()
+            from kubernetes import client, config
+            try:
+                config.load_incluster_config()
+            except Exception:  # noqa: BLE001
+                config.load_kube_config()
+            self.v1 = client.CoreV1Api()
+            self.crd = client.CustomObjectsApi()
+        except ImportError:
+            from .scheduler.kubeclient import RestCoreV1, RestCustomObjects
+            self.v1 = RestCoreV1()
+            self.crd = RestCustomObjects(self.v1)
 
     def reconcile_once(self):
         objs = self.crd.list_cluster_custom_object(
@@ -79,11 +87,13 @@ class SharePodController:
             except Exception:  # noqa: BLE001
                 self.v1.create_namespaced_pod(ns, sharepod_to_pod(obj))
                 pod = None
-            status = {
-                "phase": pod.status.phase if pod else "Creating",
-                "node": (pod.spec.node_name or "") if pod else "",
-                "podName": name,
-            }
+            phase = "Creating"
+            node = ""
+            if pod is not None:
+                phase = (pod.status.phase if pod.status else None) \
+                    or "Pending"
+                node = pod.spec.node_name or ""
+            status = {"phase": phase, "node": node, "podName": name}
             self.crd.patch_namespaced_custom_object_status(
                 self.GROUP, self.VERSION, ns, self.PLURAL, name,
                 {"status": status})

@@ -73,6 +73,8 @@ class FakeAPIServer:
         self.mutator = mutator
         self.nodes: dict[str, dict] = {}
         self.pods: dict[tuple, dict] = {}   # (ns, name) -> pod dict
+        # custom resources: (group, plural, ns, name) -> object dict
+        self.crs: dict[tuple, dict] = {}
         self.lock = threading.Lock()
         self._uid = 0
         self._now = 0
@@ -173,10 +175,38 @@ class FakeAPIServer:
                 parts = [p for p in u.path.split("/") if p]
                 return u, q, parts
 
+            # /apis/{group}/{version}/... custom-resource routing:
+            # returns (group, ns, plural, name, subresource) or None
+            def _cr_route(self, parts):
+                if len(parts) < 4 or parts[0] != "apis":
+                    return None
+                group = parts[1]
+                rest = parts[3:]
+                if rest and rest[0] == "namespaces" and len(rest) >= 3:
+                    ns, plural = rest[1], rest[2]
+                    name = rest[3] if len(rest) > 3 else None
+                    sub = rest[4] if len(rest) > 4 else None
+                    return group, ns, plural, name, sub
+                return group, None, rest[0] if rest else None, None, None
+
             def do_GET(self):
                 _, q, parts = self._route()
                 lsel = q.get("labelSelector", "")
                 fsel = q.get("fieldSelector", "")
+                cr = self._cr_route(parts)
+                if cr is not None:
+                    group, ns, plural, name, _ = cr
+                    with state.lock:
+                        if name is not None:
+                            obj = state.crs.get((group, plural, ns, name))
+                            if obj is None:
+                                return self._send({"kind": "Status",
+                                                   "code": 404}, 404)
+                            return self._send(obj)
+                        items = [o for (g, p, n, _), o in state.crs.items()
+                                 if g == group and p == plural
+                                 and (ns is None or n == ns)]
+                        return self._send({"kind": "List", "items": items})
                 with state.lock:
                     if parts[:3] == ["api", "v1", "nodes"]:
                         items = [n for n in state.nodes.values()
@@ -207,10 +237,43 @@ class FakeAPIServer:
                 _, _, parts = self._route()
                 n = int(self.headers.get("Content-Length", 0))
                 body = json.loads(self.rfile.read(n) or b"{}")
+                cr = self._cr_route(parts)
+                if cr is not None and cr[1] is not None:
+                    group, ns, plural, _, _ = cr
+                    with state.lock:
+                        state._uid += 1
+                        meta = body.setdefault("metadata", {})
+                        meta.setdefault("namespace", ns)
+                        meta.setdefault("uid", f"uid-{state._uid}")
+                        state.crs[(group, plural, ns,
+                                   meta.get("name"))] = body
+                    return self._send(body, 201)
                 if len(parts) == 5 and parts[4] == "pods":
                     ns = parts[3]
                     created = state._create_pod(ns, body)
                     return self._send(created, 201)
+                self._send({"kind": "Status", "code": 404}, 404)
+
+            def do_PATCH(self):
+                _, _, parts = self._route()
+                n = int(self.headers.get("Content-Length", 0))
+                body = json.loads(self.rfile.read(n) or b"{}")
+                cr = self._cr_route(parts)
+                if cr is not None and cr[3] is not None:
+                    group, ns, plural, name, sub = cr
+                    with state.lock:
+                        obj = state.crs.get((group, plural, ns, name))
+                        if obj is None:
+                            return self._send({"kind": "Status",
+                                               "code": 404}, 404)
+                        # merge-patch: top-level keys replace
+                        for k, v in body.items():
+                            if isinstance(v, dict) and \
+                                    isinstance(obj.get(k), dict):
+                                obj[k].update(v)
+                            else:
+                                obj[k] = v
+                        return self._send(obj)
                 self._send({"kind": "Status", "code": 404}, 404)
 
             def do_DELETE(self):

@@ -132,3 +132,35 @@ def test_resync_after_restart_over_http(cluster):
     assert leaf.available == pytest.approx(0.5)
     port = int(ann[C.POD_MANAGER_PORT])
     assert not fresh.sched.ports["node-a"].is_free(port)
+
+
+def test_sharepod_crd_end_to_end(cluster):
+    """SharePod CRD flow over HTTP: create a SharePod custom object ->
+    controller materializes the labeled Pod -> scheduler places it ->
+    controller mirrors phase/node into the SharePod status (reference
+    ships the KubeShare 1.x CRD alongside the label API, crd/v1.yaml)."""
+    from kubeshare_amd.scheduler.kubeclient import RestCustomObjects
+    from kubeshare_amd.sharepod import SharePodController
+
+    srv, api, driver = cluster
+    crd = RestCustomObjects(api)
+    G, V, P = SharePodController.GROUP, "v1", SharePodController.PLURAL
+    crd.create_namespaced_custom_object(G, V, "default", P, {
+        "apiVersion": f"{G}/v1", "kind": "SharePod",
+        "metadata": {"name": "sp1", "namespace": "default"},
+        "spec": {"gpuRequest": "0.5", "gpuLimit": "1.0", "priority": "50",
+                 "template": {"spec": {"containers": [
+                     {"name": "main", "image": "rocm/pytorch"}]}}},
+    })
+    ctl = SharePodController(v1=api, crd=crd)
+    ctl.reconcile_once()           # materialize the pod
+    pod = api.read_namespaced_pod("sp1", "default")
+    assert pod.spec.scheduler_name == C.SCHEDULER_NAME
+    labels = dict(pod.metadata.labels.items())
+    assert labels[C.POD_GPU_REQUEST] == "0.5"
+    assert labels[C.POD_PRIORITY] == "50"
+    driver.run_once()              # schedule it
+    ctl.reconcile_once()           # mirror status
+    obj = crd.get_namespaced_custom_object(G, V, "default", P, "sp1")
+    assert obj["status"]["phase"] == "Running"
+    assert obj["status"]["node"] == "node-a"
