@@ -149,6 +149,9 @@ class FakeAPIServer:
                         if ref.get("fieldPath") == "status.hostIP":
                             e["value"] = addr
                             e.pop("valueFrom", None)
+            else:
+                # real apiserver: an unscheduled pod is phase Pending
+                pod.setdefault("status", {}).setdefault("phase", "Pending")
             key = (pod["metadata"]["namespace"], pod["metadata"]["name"])
             self.pods[key] = pod
             return pod
