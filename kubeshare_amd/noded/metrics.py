@@ -35,11 +35,23 @@ class GpuSchdCollector:
         grants = CounterMetricFamily(
             "gpu_pod_token_grants",
             "cumulative token grants", labels=["node", "uuid", "pod"])
+        sampler = GaugeMetricFamily(
+            "gpu_schd_busy_sampler",
+            "1 = leases charged server-sampled GPU-busy, 0 = wall/RET",
+            labels=["node", "uuid"])
+        other = CounterMetricFamily(
+            "gpu_schd_unattributed_busy_ms",
+            "sampled GPU-busy with no token holder (exempt RCCL "
+            "kernels, ungated processes)", labels=["node", "uuid"])
         for uuid, (host, port) in self.endpoints.items():
             try:
                 st = query_stats(host, port, timeout=2.0)
             except OSError:
                 continue
+            sampler.add_metric([self.node_name, uuid],
+                               1.0 if st.get("sampler") else 0.0)
+            other.add_metric([self.node_name, uuid],
+                             st.get("other_busy_ms", 0.0))
             for pod, v in st.get("pods", {}).items():
                 lab = [self.node_name, uuid, pod]
                 usage.add_metric(lab, v["usage_ms"])
@@ -50,6 +62,8 @@ class GpuSchdCollector:
         yield share
         yield total
         yield grants
+        yield sampler
+        yield other
 
 
 def serve(endpoints: dict, node_name: str, port: int):
