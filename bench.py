@@ -72,7 +72,10 @@ def main():
     ap.add_argument("--device", default=None,
                     help="override (cpu for plumbing tests)")
     ap.add_argument("--use-ops", default="auto")
-    ap.add_argument("--fused-sgd", action="store_true")
+    # HIP multi-tensor SGD measured ~4.7% faster end-to-end than
+    # torch.optim.SGD on MI355X (same semantics, numerics-tested)
+    ap.add_argument("--fused-sgd", action=argparse.BooleanOptionalAction,
+                    default=True)
     args = ap.parse_args()
 
     import torch

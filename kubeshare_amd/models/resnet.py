@@ -36,12 +36,16 @@ class Bottleneck(nn.Module):
             )
 
     def forward(self, x):
-        identity = x if self.down is None else self.down(x)
         if self.fused_ops:
             from .. import ops
+            # downsample BN has no activation: fused no-ReLU variant
+            # (otherwise it falls back to MIOpen's fp32 spatial BN)
+            identity = x if self.down is None else \
+                ops.bn_relu(self.down[0](x), self.down[1], relu=False)
             out = ops.bn_relu(self.conv1(x), self.bn1)
             out = ops.bn_relu(self.conv2(out), self.bn2)
             return ops.bn_relu(self.conv3(out), self.bn3, res=identity)
+        identity = x if self.down is None else self.down(x)
         out = self.relu(self.bn1(self.conv1(x)))
         out = self.relu(self.bn2(self.conv2(out)))
         out = self.bn3(self.conv3(out))
@@ -68,11 +72,13 @@ class BasicBlock(nn.Module):
             )
 
     def forward(self, x):
-        identity = x if self.down is None else self.down(x)
         if self.fused_ops:
             from .. import ops
+            identity = x if self.down is None else \
+                ops.bn_relu(self.down[0](x), self.down[1], relu=False)
             out = ops.bn_relu(self.conv1(x), self.bn1)
             return ops.bn_relu(self.conv2(out), self.bn2, res=identity)
+        identity = x if self.down is None else self.down(x)
         out = self.relu(self.bn1(self.conv1(x)))
         out = self.bn2(self.conv2(out))
         return self.relu(out + identity)

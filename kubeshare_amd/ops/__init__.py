@@ -127,23 +127,25 @@ def _bn_relu_autograd():
 
         @staticmethod
         def forward(ctx, x, res, weight, bias, running_mean, running_var,
-                    training, momentum, eps):
+                    training, momentum, eps, relu):
             ext = _load()
             w32 = weight.float()
             b32 = bias.float()
             if training:
                 y, mean, invstd = ext.bn_relu_fwd_train(
                     x, w32, b32, running_mean, running_var, momentum, eps,
-                    res)
+                    res, relu)
                 ctx.save_for_backward(x, y, w32, b32, mean, invstd)
                 ctx.with_res = res is not None
+                ctx.relu = relu
                 ctx.wdtype = weight.dtype
                 return y
             y = ext.bn_relu_fwd_eval(x, w32, b32, running_mean.float(),
-                                     running_var.float(), eps, res)
+                                     running_var.float(), eps, res, relu)
             ctx.save_for_backward(x, y, w32, b32, running_mean.float(),
                                   (running_var.float() + eps).rsqrt())
             ctx.with_res = res is not None
+            ctx.relu = relu
             ctx.wdtype = weight.dtype
             return y
 
@@ -152,11 +154,11 @@ def _bn_relu_autograd():
             ext = _load()
             x, y, w32, b32, mean, invstd = ctx.saved_tensors
             out = ext.bn_relu_bwd(x, y, dy, w32, b32, mean, invstd,
-                                  ctx.with_res)
+                                  ctx.with_res, ctx.relu)
             dx, dscale, dbias = out[0], out[1], out[2]
             dres = out[3] if ctx.with_res else None
             return (dx, dres, dscale.to(ctx.wdtype), dbias.to(ctx.wdtype),
-                    None, None, None, None, None)
+                    None, None, None, None, None, None)
 
     _BNReLUFn = BNReLUFn
     return BNReLUFn
@@ -165,10 +167,11 @@ def _bn_relu_autograd():
 _BNReLUFn = None
 
 
-def bn_relu(x, bn, res=None):
-    """Fused BN+ReLU (+residual add) using an nn.BatchNorm2d's
-    parameters/buffers; falls back to eager for shapes the kernel does
-    not cover (non-bf16, C%8!=0, not channels-last)."""
+def bn_relu(x, bn, res=None, relu=True):
+    """Fused BN[+ReLU][+residual add] using an nn.BatchNorm2d's
+    parameters/buffers; relu=False serves activation-free BNs (the
+    ResNet downsample path). Falls back to eager for shapes the kernel
+    does not cover (non-bf16, C%8!=0, not channels-last)."""
     import torch
 
     supported = (x.dtype == torch.bfloat16 and x.size(1) % 8 == 0
@@ -184,13 +187,13 @@ def bn_relu(x, bn, res=None):
             bn.training, bn.momentum, bn.eps)
         if res is not None:
             y = y + res
-        return torch.relu(y)
+        return torch.relu(y) if relu else y
     if res is not None and not res.is_contiguous(
             memory_format=torch.channels_last):
         res = res.contiguous(memory_format=torch.channels_last)
     fn = _bn_relu_autograd()
     return fn.apply(x, res, bn.weight, bn.bias, bn.running_mean,
-                    bn.running_var, bn.training, bn.momentum, bn.eps)
+                    bn.running_var, bn.training, bn.momentum, bn.eps, relu)
 
 
 def fuse_model(model):

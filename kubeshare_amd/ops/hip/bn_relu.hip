@@ -186,8 +186,10 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sum,
 }
 
 // --------------------------------------------------------- fwd apply
-// y = relu(x*scale' + bias' [+ res]); scale/bias in registers.
-template <bool WITH_RES>
+// y = [relu](x*scale' + bias' [+ res]); scale/bias in registers.
+// RELU=false serves the reference ResNet's downsample-path BNs (no
+// activation), which otherwise fall back to MIOpen's fp32 spatial BN.
+template <bool WITH_RES, bool RELU = true>
 __global__ void bn_apply_relu_kernel(const ushort8* __restrict__ x,
                                      const ushort8* __restrict__ res,
                                      ushort8* __restrict__ y,
@@ -223,7 +225,7 @@ __global__ void bn_apply_relu_kernel(const ushort8* __restrict__ x,
       for (int j = 0; j < 8; j++) {
         float f = fmaf(bf16_to_f32(v[u][j]), s[j], b[j]);
         if (WITH_RES) f += bf16_to_f32(r[u][j]);
-        o[u][j] = f32_to_bf16(f > 0.f ? f : 0.f);
+        o[u][j] = f32_to_bf16(RELU && f < 0.f ? 0.f : f);
       }
       y[k[u]] = o[u];
     }
@@ -238,7 +240,7 @@ __global__ void bn_apply_relu_kernel(const ushort8* __restrict__ x,
     for (int j = 0; j < 8; j++) {
       float f = fmaf(bf16_to_f32(v[j]), s[j], b[j]);
       if (WITH_RES) f += bf16_to_f32(r[j]);
-      o[j] = f32_to_bf16(f > 0.f ? f : 0.f);
+      o[j] = f32_to_bf16(RELU && f < 0.f ? 0.f : f);
     }
     y[k] = o;
   }
@@ -345,6 +347,8 @@ __global__ void bn_bwd_stats_kernel(const ushort8* __restrict__ x,
 // per-channel constants (y>0 <=> bf16(relu(s*x+b)) != 0, replicating
 // the forward's rounding exactly), so the backward never re-reads y:
 // 7 tensor passes -> 5 for the non-residual blocks of ResNet50.
+// RELU=false (downsample BNs): no mask, g = dy.
+template <bool RELU = true>
 __global__ void bn_bwd_stats_noy_kernel(const ushort8* __restrict__ x,
                                         const ushort8* __restrict__ dy,
                                         const float* __restrict__ mean,
@@ -386,10 +390,12 @@ __global__ void bn_bwd_stats_noy_kernel(const ushort8* __restrict__ x,
 #pragma unroll
       for (int j = 0; j < 8; j++) {
         float xf = bf16_to_f32(xv[u][j]);
-        float f = fmaf(xf, sc[j], bf[j]);
-        // exact fwd replication: y stored as bf16(relu(f))
-        float g = f32_to_bf16(f > 0.f ? f : 0.f)
-                      ? bf16_to_f32(gv[u][j]) : 0.f;
+        float g = bf16_to_f32(gv[u][j]);
+        if (RELU) {
+          // exact fwd replication: y stored as bf16(relu(f))
+          float f = fmaf(xf, sc[j], bf[j]);
+          if (!f32_to_bf16(f > 0.f ? f : 0.f)) g = 0.f;
+        }
         adb[j] += g;
         ads[j] += g * ((xf - mu[j]) * is[j]);
       }
@@ -400,8 +406,11 @@ __global__ void bn_bwd_stats_noy_kernel(const ushort8* __restrict__ x,
 #pragma unroll
     for (int j = 0; j < 8; j++) {
       float xf = bf16_to_f32(xv[j]);
-      float f = fmaf(xf, sc[j], bf[j]);
-      float g = f32_to_bf16(f > 0.f ? f : 0.f) ? bf16_to_f32(gv[j]) : 0.f;
+      float g = bf16_to_f32(gv[j]);
+      if (RELU) {
+        float f = fmaf(xf, sc[j], bf[j]);
+        if (!f32_to_bf16(f > 0.f ? f : 0.f)) g = 0.f;
+      }
       adb[j] += g;
       ads[j] += g * ((xf - mu[j]) * is[j]);
     }
@@ -436,6 +445,7 @@ __global__ void bn_bwd_stats_noy_kernel(const ushort8* __restrict__ x,
 }
 
 // ----------------------------------------- bwd apply, no-y variant
+template <bool RELU = true>
 __global__ void bn_bwd_apply_noy_kernel(const ushort8* __restrict__ x,
                                         const ushort8* __restrict__ dy,
                                         ushort8* __restrict__ dx,
@@ -482,9 +492,11 @@ __global__ void bn_bwd_apply_noy_kernel(const ushort8* __restrict__ x,
 #pragma unroll
       for (int j = 0; j < 8; j++) {
         float xf = bf16_to_f32(xv[u][j]);
-        float f = fmaf(xf, sc[j], bf[j]);
-        float g = f32_to_bf16(f > 0.f ? f : 0.f)
-                      ? bf16_to_f32(gv[u][j]) : 0.f;
+        float g = bf16_to_f32(gv[u][j]);
+        if (RELU) {
+          float f = fmaf(xf, sc[j], bf[j]);
+          if (!f32_to_bf16(f > 0.f ? f : 0.f)) g = 0.f;
+        }
         float xhat = (xf - mu[j]) * is[j];
         o[j] = f32_to_bf16(w[j] * (g - db[j] - xhat * ds[j]));
       }
@@ -498,8 +510,11 @@ __global__ void bn_bwd_apply_noy_kernel(const ushort8* __restrict__ x,
 #pragma unroll
     for (int j = 0; j < 8; j++) {
       float xf = bf16_to_f32(xv[j]);
-      float f = fmaf(xf, sc[j], bf[j]);
-      float g = f32_to_bf16(f > 0.f ? f : 0.f) ? bf16_to_f32(gv[j]) : 0.f;
+      float g = bf16_to_f32(gv[j]);
+      if (RELU) {
+        float f = fmaf(xf, sc[j], bf[j]);
+        if (!f32_to_bf16(f > 0.f ? f : 0.f)) g = 0.f;
+      }
       float xhat = (xf - mu[j]) * is[j];
       o[j] = f32_to_bf16(w[j] * (g - db[j] - xhat * ds[j]));
     }
@@ -688,7 +703,7 @@ Geom geom_of(const torch::Tensor& x) {
 std::vector<torch::Tensor> bn_relu_fwd_train(
     torch::Tensor x, torch::Tensor weight, torch::Tensor bias,
     torch::Tensor running_mean, torch::Tensor running_var, double momentum,
-    double eps, c10::optional<torch::Tensor> res) {
+    double eps, c10::optional<torch::Tensor> res, bool relu) {
   Geom g = geom_of(x);
   auto stream = at::cuda::getCurrentCUDAStream();
   auto f32 = x.options().dtype(at::kFloat);
@@ -721,29 +736,30 @@ std::vector<torch::Tensor> bn_relu_fwd_train(
                      save_invstd.data_ptr<float>(), scale.data_ptr<float>(),
                      biasf.data_ptr<float>(), g.M, g.C, (float)momentum,
                      (float)eps);
-  if (res.has_value()) {
-    hipLaunchKernelGGL(bn_apply_relu_kernel<true>, dim3(g.blocks),
-                       dim3(KS_BN_BLOCK), 0, stream.stream(),
+  auto launch_apply = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(g.blocks), dim3(KS_BN_BLOCK), 0,
+                       stream.stream(),
                        reinterpret_cast<const ushort8*>(x.data_ptr()),
-                       reinterpret_cast<const ushort8*>(res->data_ptr()),
+                       res.has_value()
+                           ? reinterpret_cast<const ushort8*>(res->data_ptr())
+                           : nullptr,
                        reinterpret_cast<ushort8*>(y.data_ptr()),
                        scale.data_ptr<float>(), biasf.data_ptr<float>(), g.M,
                        g.CG);
-  } else {
-    hipLaunchKernelGGL(bn_apply_relu_kernel<false>, dim3(g.blocks),
-                       dim3(KS_BN_BLOCK), 0, stream.stream(),
-                       reinterpret_cast<const ushort8*>(x.data_ptr()),
-                       nullptr, reinterpret_cast<ushort8*>(y.data_ptr()),
-                       scale.data_ptr<float>(), biasf.data_ptr<float>(), g.M,
-                       g.CG);
-  }
+  };
+  if (res.has_value())
+    relu ? launch_apply(bn_apply_relu_kernel<true, true>)
+         : launch_apply(bn_apply_relu_kernel<true, false>);
+  else
+    relu ? launch_apply(bn_apply_relu_kernel<false, true>)
+         : launch_apply(bn_apply_relu_kernel<false, false>);
   return {y, save_mean, save_invstd};
 }
 
 torch::Tensor bn_relu_fwd_eval(torch::Tensor x, torch::Tensor weight,
                                torch::Tensor bias, torch::Tensor rmean,
                                torch::Tensor rvar, double eps,
-                               c10::optional<torch::Tensor> res) {
+                               c10::optional<torch::Tensor> res, bool relu) {
   Geom g = geom_of(x);
   auto stream = at::cuda::getCurrentCUDAStream();
   auto f32 = x.options().dtype(at::kFloat);
@@ -756,29 +772,33 @@ torch::Tensor bn_relu_fwd_eval(torch::Tensor x, torch::Tensor weight,
                      bias.data_ptr<float>(), rmean.data_ptr<float>(),
                      rvar.data_ptr<float>(), scale.data_ptr<float>(),
                      biasf.data_ptr<float>(), g.C, (float)eps);
-  if (res.has_value()) {
-    hipLaunchKernelGGL(bn_apply_relu_kernel<true>, dim3(g.blocks),
-                       dim3(KS_BN_BLOCK), 0, stream.stream(),
+  auto launch_apply = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(g.blocks), dim3(KS_BN_BLOCK), 0,
+                       stream.stream(),
                        reinterpret_cast<const ushort8*>(x.data_ptr()),
-                       reinterpret_cast<const ushort8*>(res->data_ptr()),
+                       res.has_value()
+                           ? reinterpret_cast<const ushort8*>(res->data_ptr())
+                           : nullptr,
                        reinterpret_cast<ushort8*>(y.data_ptr()),
                        scale.data_ptr<float>(), biasf.data_ptr<float>(), g.M,
                        g.CG);
-  } else {
-    hipLaunchKernelGGL(bn_apply_relu_kernel<false>, dim3(g.blocks),
-                       dim3(KS_BN_BLOCK), 0, stream.stream(),
-                       reinterpret_cast<const ushort8*>(x.data_ptr()),
-                       nullptr, reinterpret_cast<ushort8*>(y.data_ptr()),
-                       scale.data_ptr<float>(), biasf.data_ptr<float>(), g.M,
-                       g.CG);
-  }
+  };
+  if (res.has_value())
+    relu ? launch_apply(bn_apply_relu_kernel<true, true>)
+         : launch_apply(bn_apply_relu_kernel<true, false>);
+  else
+    relu ? launch_apply(bn_apply_relu_kernel<false, true>)
+         : launch_apply(bn_apply_relu_kernel<false, false>);
   return y;
 }
 
 std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor x, torch::Tensor y,
                                        torch::Tensor dy, torch::Tensor weight,
                                        torch::Tensor bias, torch::Tensor mean,
-                                       torch::Tensor invstd, bool need_dres) {
+                                       torch::Tensor invstd, bool need_dres,
+                                       bool relu) {
+  TORCH_CHECK(relu || !need_dres,
+              "bn_relu_bwd: residual path requires the ReLU variant");
   Geom g = geom_of(x);
   auto stream = at::cuda::getCurrentCUDAStream();
   auto f32 = x.options().dtype(at::kFloat);
@@ -821,25 +841,33 @@ std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor x, torch::Tensor y,
   }
   // non-residual: the ReLU mask is recomputed from x and the folded
   // scale/bias, so y is never read (7 tensor passes -> 5)
-  hipLaunchKernelGGL(bn_bwd_stats_noy_kernel, dim3(g.stat_blocks),
-                     dim3(KS_BN_BLOCK), 0, stream.stream(),
-                     reinterpret_cast<const ushort8*>(x.data_ptr()),
-                     reinterpret_cast<const ushort8*>(dy.data_ptr()),
-                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                     weight.data_ptr<float>(), bias.data_ptr<float>(),
-                     part.data_ptr<float>(), g.M, g.CG);
+  auto launch_stats = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(g.stat_blocks), dim3(KS_BN_BLOCK), 0,
+                       stream.stream(),
+                       reinterpret_cast<const ushort8*>(x.data_ptr()),
+                       reinterpret_cast<const ushort8*>(dy.data_ptr()),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       weight.data_ptr<float>(), bias.data_ptr<float>(),
+                       part.data_ptr<float>(), g.M, g.CG);
+  };
+  relu ? launch_stats(bn_bwd_stats_noy_kernel<true>)
+       : launch_stats(bn_bwd_stats_noy_kernel<false>);
   hipLaunchKernelGGL(bn_reduce_partials_kernel,
                      dim3((g.C + 3) / 4), dim3(64, 4), 0, stream.stream(),
                      part.data_ptr<float>(), g.stat_blocks, g.C,
                      dbias_p, dscale_p);
-  hipLaunchKernelGGL(bn_bwd_apply_noy_kernel, dim3(g.blocks),
-                     dim3(KS_BN_BLOCK), 0, stream.stream(),
-                     reinterpret_cast<const ushort8*>(x.data_ptr()),
-                     reinterpret_cast<const ushort8*>(dy.data_ptr()),
-                     reinterpret_cast<ushort8*>(dx.data_ptr()),
-                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                     weight.data_ptr<float>(), bias.data_ptr<float>(),
-                     dbias_p, dscale_p, g.M, g.CG);
+  auto launch_bapply = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(g.blocks), dim3(KS_BN_BLOCK), 0,
+                       stream.stream(),
+                       reinterpret_cast<const ushort8*>(x.data_ptr()),
+                       reinterpret_cast<const ushort8*>(dy.data_ptr()),
+                       reinterpret_cast<ushort8*>(dx.data_ptr()),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       weight.data_ptr<float>(), bias.data_ptr<float>(),
+                       dbias_p, dscale_p, g.M, g.CG);
+  };
+  relu ? launch_bapply(bn_bwd_apply_noy_kernel<true>)
+       : launch_bapply(bn_bwd_apply_noy_kernel<false>);
   auto dbias = red.narrow(0, 0, g.C);
   auto dscale = red.narrow(0, g.C, g.C);
   return {dx, dscale, dbias};

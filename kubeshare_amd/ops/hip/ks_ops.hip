@@ -122,15 +122,16 @@ void sgd_momentum_(std::vector<torch::Tensor> params,
 std::vector<torch::Tensor> bn_relu_fwd_train(
     torch::Tensor x, torch::Tensor weight, torch::Tensor bias,
     torch::Tensor running_mean, torch::Tensor running_var, double momentum,
-    double eps, c10::optional<torch::Tensor> res);
+    double eps, c10::optional<torch::Tensor> res, bool relu);
 torch::Tensor bn_relu_fwd_eval(torch::Tensor x, torch::Tensor weight,
                                torch::Tensor bias, torch::Tensor rmean,
                                torch::Tensor rvar, double eps,
-                               c10::optional<torch::Tensor> res);
+                               c10::optional<torch::Tensor> res, bool relu);
 std::vector<torch::Tensor> bn_relu_bwd(torch::Tensor x, torch::Tensor y,
                                        torch::Tensor dy, torch::Tensor weight,
                                        torch::Tensor bias, torch::Tensor mean,
-                                       torch::Tensor invstd, bool need_dres);
+                                       torch::Tensor invstd, bool need_dres,
+                                       bool relu);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("burn", &burn, "occupy the GPU for ~ms milliseconds",

@@ -209,3 +209,36 @@ def test_bn_relu_odd_channels_falls_back():
     y = ops.bn_relu(x, bn)
     assert y.shape == x.shape
     assert (y.float() >= 0).all()
+
+
+def test_bn_only_no_relu_matches_torch():
+    """relu=False variant (downsample-path BN, no activation) vs plain
+    fp32 torch reference — forward and all gradients."""
+    shape = (16, 64, 28, 28)
+    x = _mk(shape).requires_grad_()
+    bn = torch.nn.BatchNorm2d(shape[1]).cuda()
+    with torch.no_grad():
+        bn.weight.uniform_(0.5, 1.5)
+        bn.bias.uniform_(-0.5, 0.5)
+    x32 = x.detach().float().requires_grad_()
+    w32 = bn.weight.detach().clone().requires_grad_()
+    b32 = bn.bias.detach().clone().requires_grad_()
+    rm, rv = bn.running_mean.clone(), bn.running_var.clone()
+    y_ref = torch.nn.functional.batch_norm(x32, rm, rv, w32, b32,
+                                           True, 0.1, 1e-5)
+    gb = torch.randn_like(y_ref).to(torch.bfloat16).contiguous(
+        memory_format=torch.channels_last)
+    y_ref.backward(gb.float())
+
+    y = ops.bn_relu(x, bn, relu=False)
+    y.backward(gb)
+    torch.cuda.synchronize()
+    assert (y.float() < 0).any(), "no-relu output must keep negatives"
+    torch.testing.assert_close(y.float(), y_ref, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(x.grad.float(), x32.grad,
+                               rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(bn.weight.grad.float(), w32.grad,
+                               rtol=2e-2, atol=2e-1)
+    torch.testing.assert_close(bn.bias.grad.float(), b32.grad,
+                               rtol=2e-2, atol=2e-1)
+    torch.testing.assert_close(bn.running_mean, rm, rtol=1e-3, atol=1e-3)

@@ -93,7 +93,7 @@ def main():
 
         def ext_fwd():
             ext.bn_relu_fwd_train(x, w32, b32, bn2.running_mean,
-                                  bn2.running_var, 0.1, 1e-5, res)
+                                  bn2.running_var, 0.1, 1e-5, res, True)
 
         def aten_fwd():
             torch.ops.aten.miopen_batch_norm(
