@@ -331,3 +331,27 @@ def test_rccl_communicator_under_sharing(share):
     finally:
         share.remove_pod("gpu/rcclrank")
         share.remove_pod("gpu/competitor")
+
+
+def test_three_pods_mixed_split(share):
+    """Config #3's quota matrix on one GPU: 0.5 + 0.25 + 0.25 requests
+    at limit=request (hard caps) must converge to ~50/25/25 busy
+    shares server-side."""
+    a = share.add_pod("gpu/half", request=0.5, limit=0.5)
+    b = share.add_pod("gpu/q1", request=0.25, limit=0.25)
+    c = share.add_pod("gpu/q2", request=0.25, limit=0.25)
+    try:
+        ps = [_spawn_burner(h, 9000, wait_go=True) for h in (a, b, c)]
+        _start_together(ps)
+        time.sleep(7.0)
+        mid = share.stats()
+        for p in ps:
+            p.communicate(timeout=120)
+        shares = {k: mid["pods"][k]["busy_share"]
+                  for k in ("gpu/half", "gpu/q1", "gpu/q2")}
+        assert abs(shares["gpu/half"] - 0.50) < 0.12, shares
+        assert abs(shares["gpu/q1"] - 0.25) < 0.10, shares
+        assert abs(shares["gpu/q2"] - 0.25) < 0.10, shares
+    finally:
+        for n in ("gpu/half", "gpu/q1", "gpu/q2"):
+            share.remove_pod(n)
