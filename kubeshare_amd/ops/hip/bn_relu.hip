@@ -48,7 +48,7 @@ __device__ __forceinline__ unsigned short f32_to_bf16(float f) {
   return (unsigned short)(v.i >> 16);
 }
 
-#define KS_BN_BLOCK 256
+#define KS_BN_BLOCK 512
 #define KS_BN_UNROLL 4
 #define KS_BN_UNROLL_STATS 8
 
@@ -686,7 +686,7 @@ Geom geom_of(const torch::Tensor& x) {
   Geom g;
   g.C = (int)x.size(1);
   TORCH_CHECK(g.C % 8 == 0 && g.C <= KS_BN_BLOCK * 8,
-              "bn_relu: C must be a multiple of 8 and <= 2048");
+              "bn_relu: C must be a multiple of 8 and <= 8*KS_BN_BLOCK");
   g.M = x.numel() / g.C;
   g.CG = g.C / 8;
   long long work = g.M * g.CG;
