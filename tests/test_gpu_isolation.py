@@ -355,3 +355,36 @@ def test_three_pods_mixed_split(share):
     finally:
         for n in ("gpu/half", "gpu/q1", "gpu/q2"):
             share.remove_pod(n)
+
+
+def test_dataloader_fork_safety(share):
+    """Real pods run torch DataLoader worker processes, which FORK the
+    hooked process: the gate must reset in the children (pthread_atfork
+    — a shared socket fd would corrupt the token stream) and training
+    must proceed gated in the parent."""
+    h = share.add_pod("gpu/loader", request=0.5, limit=1.0)
+    code = (
+        "import torch, ctypes\n"
+        "from torch.utils.data import DataLoader, TensorDataset\n"
+        "ds = TensorDataset(torch.randn(64, 3, 32, 32),\n"
+        "                   torch.randint(0, 10, (64,)))\n"
+        "dl = DataLoader(ds, batch_size=16, num_workers=2)\n"
+        "m = torch.nn.Conv2d(3, 8, 3).cuda()\n"
+        "opt = torch.optim.SGD(m.parameters(), lr=0.1)\n"
+        "for x, y in dl:\n"
+        "    loss = m(x.cuda()).square().mean()\n"
+        "    opt.zero_grad(); loss.backward(); opt.step()\n"
+        "torch.cuda.synchronize()\n"
+        "lib = ctypes.CDLL(None)\n"
+        "lib.ks_hook_leases.restype = ctypes.c_longlong\n"
+        "assert lib.ks_hook_leases() >= 1, 'gate never engaged'\n"
+        "print('LOADER-OK')\n"
+    )
+    try:
+        env = h.env(gpu_index=0)
+        env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+        r = subprocess.run([sys.executable, "-c", code], env=env, cwd=REPO,
+                           capture_output=True, text=True, timeout=300)
+        assert "LOADER-OK" in r.stdout, (r.stdout[-2000:], r.stderr[-2000:])
+    finally:
+        share.remove_pod("gpu/loader")
