@@ -388,3 +388,22 @@ def test_dataloader_fork_safety(share):
         assert "LOADER-OK" in r.stdout, (r.stdout[-2000:], r.stderr[-2000:])
     finally:
         share.remove_pod("gpu/loader")
+
+
+def test_serving_under_sharing_latency(native_bins):
+    """Serving story: a hipGraph-replay inference pod co-located with a
+    saturating trainer. With a latency-oriented base quota (-q 50) the
+    server's p99 must stay bounded by a few lease lengths — the
+    time-slicing latency floor — while the trainer keeps the bulk of
+    the GPU."""
+    sys.path.insert(0, REPO)
+    from tools.serve_probe import run_config
+    r = run_config(quota_ms=50.0, duration_ms=8000)
+    lat = r["latency"]
+    assert lat["n"] >= 50
+    # floor: request waits for the trainer's lease drain; a few leases
+    # of slack for scheduling + graph replay itself
+    assert lat["p99_ms"] < 50.0 * 6, lat
+    assert lat["p50_ms"] < 50.0 * 4, lat
+    # the trainer still gets most of the GPU
+    assert r["trainer_busy_frac"] > 0.5, r
