@@ -62,6 +62,8 @@ class GraphReplayServer:
             self.serve(x)
             lat.append((time.perf_counter() - t0) * 1000.0)
             time.sleep(interarrival_s)
+        outliers = [(i, round(v, 1)) for i, v in enumerate(lat)
+                    if v > 200.0]
         lat.sort()
 
         def pct(p):
@@ -69,4 +71,5 @@ class GraphReplayServer:
         return {"n": len(lat), "p50_ms": round(pct(50), 2),
                 "p95_ms": round(pct(95), 2), "p99_ms": round(pct(99), 2),
                 "max_ms": round(lat[-1], 2),
-                "mean_ms": round(sum(lat) / len(lat), 2)}
+                "mean_ms": round(sum(lat) / len(lat), 2),
+                "outliers": outliers[:10]}

@@ -175,6 +175,8 @@ class TokenScheduler {
         account(it->first).charge(
             now, sampler_active_ ? it->second.sampled_ms
                                  : now - it->second.grant_ms);
+        last_revoked_ = it->first;
+        revokes_++;
         it = holders_.erase(it);
         any = true;
       } else {
@@ -329,6 +331,8 @@ class TokenScheduler {
                     ",\"busy_ms\":" + fmt(total_busy) +
                     ",\"sampler\":" + (sampler_active_ ? "true" : "false") +
                     ",\"other_busy_ms\":" + fmt(other_busy_ms_) +
+                    ",\"revokes\":" + std::to_string(revokes_) +
+                    ",\"last_revoked\":\"" + last_revoked_ + "\"" +
                     ",\"pods\":{";
     bool first = true;
     for (auto& kv : pods_) {
@@ -370,6 +374,8 @@ class TokenScheduler {
   std::string holder_group_;  // group of the current holders ("": none)
   bool sampler_active_ = false;
   double other_busy_ms_ = 0.0;  // sampled busy with no holder (stats)
+  long long revokes_ = 0;       // liveness revocations (lost RETs)
+  std::string last_revoked_;
 };
 
 }  // namespace ks

@@ -76,6 +76,8 @@ def run_config(quota_ms: float, duration_ms: float) -> dict:
         wall = float(out_t.split()[1])
         return {"quota_ms": quota_ms, "latency": served,
                 "trainer_busy_frac": round(trainer_used / (wall * 1000), 3),
+                "revokes": st.get("revokes"),
+                "last_revoked": st.get("last_revoked"),
                 "schd": {k: round(v["busy_share"], 3)
                          for k, v in st.get("pods", {}).items()}}
     finally:
