@@ -56,14 +56,23 @@ class GraphReplayServer:
 
     def bench(self, x, n_requests: int = 100,
               interarrival_s: float = 0.05) -> dict:
-        lat = []
+        import ctypes
+        try:
+            lib = ctypes.CDLL(None)
+            lib.ks_hook_wait_ms.restype = ctypes.c_double
+            wait_ms = lib.ks_hook_wait_ms
+        except (OSError, AttributeError):
+            wait_ms = lambda: 0.0  # noqa: E731 — hook not attached
+        lat, waits = [], []
         for _ in range(n_requests):
+            w0 = wait_ms()
             t0 = time.perf_counter()
             self.serve(x)
             lat.append((time.perf_counter() - t0) * 1000.0)
+            waits.append(wait_ms() - w0)
             time.sleep(interarrival_s)
-        outliers = [(i, round(v, 1)) for i, v in enumerate(lat)
-                    if v > 200.0]
+        outliers = [(i, round(v, 1), round(waits[i], 1))
+                    for i, v in enumerate(lat) if v > 200.0]
         lat.sort()
 
         def pct(p):
