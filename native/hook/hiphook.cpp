@@ -687,6 +687,7 @@ int ks_hook_active(void) {
 }
 long long ks_hook_leases(void) { return S().gate.leases(); }
 double ks_hook_used_ms(void) { return S().gate.used_ms_total(); }
+double ks_hook_wait_ms(void) { return S().gate.acquire_wait_ms_total(); }
 long long ks_hook_mem_used(void) {
   HookState& s = S();
   std::lock_guard<std::mutex> g(s.mem_mu);

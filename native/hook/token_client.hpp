@@ -192,6 +192,7 @@ class TokenGate {
 
   long long leases() const { return client_.requests(); }
   double used_ms_total() const { return client_.returned_ms(); }
+  double acquire_wait_ms_total() const { return acquire_wait_ms_; }
 
  private:
   static void* watchdog_entry(void* self) {
@@ -271,10 +272,12 @@ class TokenGate {
       holding_ = false;
     }
     start_watchdog_locked();
+    double t_wait0 = now_ms();
     // hint = EWMA of recent per-lease use, so gpu-schd can right-size
     // the next quota (a bursty pod gets shorter leases; see
     // token_sched.hpp schedule()). KUBESHARE_LEASE_HINT=0 disables.
     double quota = client_.acquire(hint_enabled_ ? ewma_used_ : 0.0);
+    acquire_wait_ms_ += now_ms() - t_wait0;
     if (quota <= 0.0) {
       // Scheduler unreachable: fail OPEN for liveness (the node daemon
       // restarts the chain; isolation degrades, jobs don't die) unless
@@ -301,6 +304,7 @@ class TokenGate {
   std::atomic<double> deadline_{-1.0};
   std::atomic<double> last_activity_{0.0};
   double idle_release_ms_ = 25.0;
+  double acquire_wait_ms_ = 0.0;  // cumulative blocked-in-REQ time
   bool holding_ = false;
   bool hint_enabled_ = true;
   double grant_time_ = 0.0;
