@@ -407,3 +407,34 @@ def test_serving_under_sharing_latency(native_bins):
     assert lat["p50_ms"] < 50.0 * 4, lat
     # the trainer still gets most of the GPU
     assert r["trainer_busy_frac"] > 0.5, r
+
+
+def test_pod_churn_under_load(share):
+    """Production churn on hardware: pods join and leave while others
+    keep the GPU busy — the chain (config rewrite + pod-mgr lifecycle +
+    token scheduling) must stay live and keep granting."""
+    a = share.add_pod("gpu/stay", request=0.5, limit=1.0)
+    b = share.add_pod("gpu/leave", request=0.5, limit=1.0)
+    try:
+        pa = _spawn_burner(a, 10000, wait_go=True)
+        pb = _spawn_burner(b, 3000, wait_go=True)
+        _start_together([pa, pb])
+        out_b, _ = pb.communicate(timeout=120)   # b finishes early
+        assert pb.returncode == 0, out_b
+        share.remove_pod("gpu/leave")            # churn: b leaves...
+        c = share.add_pod("gpu/join", request=0.25, limit=1.0)
+        pc = _spawn_burner(c, 3000, wait_go=True)
+        assert pc.stdout.readline().strip() == "READY"
+        pc.stdin.write("GO\n")
+        pc.stdin.flush()                         # ...c joins mid-run
+        out_c, _ = pc.communicate(timeout=120)
+        out_a, _ = pa.communicate(timeout=120)
+        assert pa.returncode == 0 and pc.returncode == 0, (out_a, out_c)
+        st = share.stats()
+        assert st["pods"]["gpu/stay"]["grants"] >= 2
+        assert st["pods"]["gpu/join"]["grants"] >= 1
+        assert float(out_a.split()[4]) > 0       # a kept making progress
+    finally:
+        for n in ("gpu/stay", "gpu/join"):
+            if n in share.pods:
+                share.remove_pod(n)
