@@ -193,6 +193,25 @@ class NodeDaemon:
                 continue
             sup.reconcile(entries)
 
+    def publish_inventory(self, node_name: str = "", api=None) -> str:
+        """Patch this Node's kubeshare.amd/gpus annotation (inventory +
+        xGMI link graph) so the cluster scheduler can consume it without
+        a Prometheus round-trip. Also labels the node SharedGPU=true."""
+        from ..scheduler.inventory import GPUInfo, format_node_annotation
+        infos = [g if isinstance(g, GPUInfo) else GPUInfo(**g)
+                 for g in self.gpus]
+        ann = format_node_annotation(infos)
+        node = node_name or os.uname().nodename
+        if api is None:
+            from ..scheduler.kube import make_client
+            api = make_client()
+        patch = {"metadata": {"annotations": {"kubeshare.amd/gpus": ann},
+                              "labels": {"SharedGPU": "true"}}}
+        # both the official CoreV1Api and the stdlib RestCoreV1 expose
+        # patch_node(name, body)
+        api.patch_node(node, patch)
+        return ann
+
     def serve_metrics(self, port: int, node_name: str = ""):
         """Export every local gpu-schd's STATS as Prometheus metrics."""
         from .metrics import serve

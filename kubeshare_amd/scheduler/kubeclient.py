@@ -217,6 +217,11 @@ class RestCoreV1:
         return self._merge_patch(
             f"/api/v1/namespaces/{namespace}/pods/{name}", body)
 
+    def patch_node(self, name: str, body: dict):
+        """Merge-patch a Node (the noded publishes its GPU inventory as
+        the kubeshare.amd/gpus annotation this way)."""
+        return self._merge_patch(f"/api/v1/nodes/{name}", body)
+
     def _merge_patch(self, path: str, body: dict):
         data = json.dumps(body).encode()
         req = urllib.request.Request(self.base_url + path, data=data,

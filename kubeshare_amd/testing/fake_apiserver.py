@@ -261,6 +261,20 @@ class FakeAPIServer:
                 _, _, parts = self._route()
                 n = int(self.headers.get("Content-Length", 0))
                 body = json.loads(self.rfile.read(n) or b"{}")
+                # /api/v1/nodes/<name> merge patch (inventory annotation)
+                if parts[:3] == ["api", "v1", "nodes"] and len(parts) == 4:
+                    with state.lock:
+                        node = state.nodes.get(parts[3])
+                        if node is None:
+                            return self._send({"kind": "Status",
+                                               "code": 404}, 404)
+                        meta = body.get("metadata", {})
+                        node.setdefault("metadata", {}).setdefault(
+                            "annotations", {}).update(
+                            meta.get("annotations") or {})
+                        node["metadata"].setdefault("labels", {}).update(
+                            meta.get("labels") or {})
+                        return self._send(node)
                 cr = self._cr_route(parts)
                 if cr is not None and cr[3] is not None:
                     group, ns, plural, name, sub = cr

@@ -164,3 +164,30 @@ def test_sharepod_crd_end_to_end(cluster):
     obj = crd.get_namespaced_custom_object(G, V, "default", P, "sp1")
     assert obj["status"]["phase"] == "Running"
     assert obj["status"]["node"] == "node-a"
+
+
+def test_noded_publishes_inventory_annotation(cluster, tmp_path):
+    """The node daemon closes the inventory loop: amdsmi inventory ->
+    kubeshare.amd/gpus node annotation (with the xGMI link graph) ->
+    KubeDriver.sync_nodes, no Prometheus round-trip."""
+    from kubeshare_amd.noded.launcher import NodeDaemon
+    from kubeshare_amd.scheduler.kube import KubeDriver
+    from kubeshare_amd.scheduler.topology import TopologyConfig
+
+    srv, api, driver = cluster
+    srv.add_node("node-inv", ready=True)  # fresh unlabeled node
+    gpus = [{"uuid": f"GPU-inv-{i}", "model": C.MI355X_MODEL,
+             "memory": C.MI355X_HBM_BYTES, "index": i,
+             "xgmi_links": {j: 1 for j in range(2) if j != i}}
+            for i in range(2)]
+    nd = NodeDaemon(str(tmp_path), gpus=gpus)
+    ann = nd.publish_inventory(node_name="node-inv", api=api)
+    assert "GPU-inv-0" in ann and "links=" in ann
+    # a driver with NO injected provider now sees the node via the
+    # annotation (SharedGPU=true label was patched too)
+    topo = TopologyConfig.single_node("node-inv", gpus=2)
+    d2 = KubeDriver(topo, inventory=None, api=api)
+    d2.sync_nodes()
+    leaf = d2.sched.tree.leaf_by_uuid["GPU-inv-0"]
+    assert leaf.full_memory == C.MI355X_HBM_BYTES
+    assert "GPU-inv-1" in (leaf.xgmi_peers or {})
