@@ -246,6 +246,10 @@ def main():
                     choices=["auto", "amdsmi", "torch"])
     ap.add_argument("--metrics-port", type=int, default=0,
                     help="serve per-pod quota-enforcement metrics")
+    ap.add_argument("--publish-inventory", action="store_true",
+                    help="patch this Node's kubeshare.amd/gpus "
+                         "annotation + SharedGPU label at startup")
+    ap.add_argument("--node-name", default=os.environ.get("NODE_NAME", ""))
     args = ap.parse_args()
     daemon = NodeDaemon(args.workdir, args.base_port, args.base_quota,
                         args.min_quota, args.window, args.inventory)
@@ -253,6 +257,11 @@ def main():
         log("no GPUs found; exiting")
         return 1
     daemon.start()
+    if args.publish_inventory:
+        try:
+            daemon.publish_inventory(args.node_name)
+        except Exception as e:  # noqa: BLE001 — annotation is best-effort
+            log(f"inventory publish failed: {e}")
     if args.metrics_port:
         daemon.serve_metrics(args.metrics_port)
     try:

@@ -89,6 +89,20 @@ class TopologyConfig:
             cells=[CellSpec(cell_type="MI355X-NODE", cell_id=node)],
         )
 
+    @classmethod
+    def nodes(cls, node_names, gpus: int = 8,
+              model: str = "AMD Instinct MI355X",
+              priority: int = 100) -> "TopologyConfig":
+        """Convenience: a flat multi-node MI355X cluster (one node-level
+        cell per host; cross-node locality uses the hierarchical cell-ID
+        distance)."""
+        return cls(
+            cell_types={
+                "MI355X-NODE": CellTypeSpec(model, gpus, priority, True)},
+            cells=[CellSpec(cell_type="MI355X-NODE", cell_id=n)
+                   for n in node_names],
+        )
+
 
 @dataclass
 class CellElement:

@@ -191,3 +191,43 @@ def test_noded_publishes_inventory_annotation(cluster, tmp_path):
     leaf = d2.sched.tree.leaf_by_uuid["GPU-inv-0"]
     assert leaf.full_memory == C.MI355X_HBM_BYTES
     assert "GPU-inv-1" in (leaf.xgmi_peers or {})
+
+
+def test_multi_node_health_and_gang_over_http(tmp_path):
+    """Two-node cluster over HTTP: an unhealthy node is avoided, a
+    4-member gang lands atomically on the healthy node, and a health
+    recovery makes the second node usable again."""
+    from kubeshare_amd.scheduler.topology import TopologyConfig
+    srv = FakeAPIServer()
+    srv.add_node("n1", labels={"SharedGPU": "true"}, host_ip="10.0.0.1")
+    srv.add_node("n2", labels={"SharedGPU": "true"}, host_ip="10.0.0.2",
+                 ready=False)
+    port = srv.start()
+    api = RestCoreV1(f"http://127.0.0.1:{port}")
+    inv = FakeInventory({"n1": {"gpus": 4}, "n2": {"gpus": 4}})
+    driver = KubeDriver(TopologyConfig.nodes(["n1", "n2"], gpus=4),
+                        inventory=inv, api=api)
+    driver.sync_nodes()
+    try:
+        labels = {C.POD_GPU_REQUEST: "1.0", C.POD_GPU_LIMIT: "1.0",
+                  C.POD_GROUP_NAME: "g4", C.POD_GROUP_HEADCOUNT: "4",
+                  C.POD_GROUP_THRESHOLD: "1.0"}
+        for i in range(4):
+            srv.submit_pod("default", f"g4-{i}", labels)
+        driver.run_once()
+        nodes = {api.read_namespaced_pod(f"g4-{i}",
+                                         "default").spec.node_name
+                 for i in range(4)}
+        assert nodes == {"n1"}  # n2 unhealthy: whole gang on n1
+        # n2 recovers; a new pod can land there (n1 is full)
+        with srv.lock:
+            srv.nodes["n2"]["status"]["conditions"][0]["status"] = "True"
+        driver.sync_nodes()
+        srv.submit_pod("default", "after",
+                       {C.POD_GPU_REQUEST: "0.5",
+                        C.POD_GPU_LIMIT: "1.0"})
+        driver.run_once()
+        pod = api.read_namespaced_pod("after", "default")
+        assert pod.spec.node_name == "n2"
+    finally:
+        srv.stop()
