@@ -69,6 +69,9 @@ def main():
     ap.add_argument("--request", type=float, default=0.5)
     ap.add_argument("--limit", type=float, default=1.0)
     ap.add_argument("--dtype", default="bf16")
+    ap.add_argument("--quota-ms", type=float, default=None,
+                    help="override gpu-schd base quota (-q); default "
+                         "keeps the reference's 300 ms")
     ap.add_argument("--device", default=None,
                     help="override (cpu for plumbing tests)")
     ap.add_argument("--use-ops", default="auto")
@@ -103,6 +106,8 @@ def main():
                           full_memory=(torch.cuda.get_device_properties(
                               local_rank).total_memory if on_gpu
                               else 288 * 1024**3))
+    if args.quota_ms:
+        share.base_quota_ms = args.quota_ms
     share.start()
 
     workers = []
