@@ -104,9 +104,11 @@ def build_patch(pod: dict) -> list:
             patch.append({"op": "add", "path": f"/spec/containers/{i}/env",
                           "value": env})
         else:
+            replaced = set()
             for e in env:
                 j = conflict_idx.get(e["name"])
                 if j is not None:
+                    replaced.add(e["name"])
                     patch.append({"op": "replace",
                                   "path": f"/spec/containers/{i}/env/{j}",
                                   "value": e})
@@ -114,6 +116,14 @@ def build_patch(pod: dict) -> list:
                     patch.append({"op": "add",
                                   "path": f"/spec/containers/{i}/env/-",
                                   "value": e})
+            # conflicting names we do NOT inject ourselves (a user-set
+            # HIP_VISIBLE_DEVICES would filter against the ROCR-pinned
+            # view and hide the device): neutralize in place
+            for name, j in conflict_idx.items():
+                if name not in replaced:
+                    patch.append({"op": "replace",
+                                  "path": f"/spec/containers/{i}/env/{j}",
+                                  "value": {"name": name, "value": "0"}})
         if shared:
             mounts = container.get("volumeMounts")
             have = {m.get("name") for m in mounts or []}

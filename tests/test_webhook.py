@@ -122,3 +122,33 @@ def test_fastapi_endpoint():
         "request": {"uid": "u", "object": _pod_with_annotations(ann)}})
     assert r.status_code == 200
     assert r.json()["response"]["allowed"] is True
+
+
+def test_patch_application_is_idempotent():
+    """Invariant: applying the webhook patch once makes a second
+    invocation a no-op (re-invocation-safe MutatingWebhookConfiguration
+    with reinvocationPolicy: IfNeeded)."""
+    from kubeshare_amd.testing.fake_apiserver import apply_json_patch
+    ann = {C.POD_GPU_UUID: "GPU-1", C.POD_MANAGER_PORT: "50055",
+           C.POD_GPU_MEMORY: "123", C.POD_GPU_INDEX: "2"}
+    pod = _pod_with_annotations(ann)
+    pod["spec"]["containers"].append({"name": "sidecar", "env": [
+        {"name": "HIP_VISIBLE_DEVICES", "value": "7"}]})
+    patch1 = build_patch(pod)
+    assert patch1
+    mutated = apply_json_patch(pod, patch1)
+    assert build_patch(mutated) == []
+    # every container got the marker + ROCR pinning
+    for c in mutated["spec"]["containers"]:
+        names = {e["name"] for e in c["env"]}
+        assert C.ENV_INJECTED in names
+        assert C.ENV_ROCR_VISIBLE_DEVICES in names
+    # the sidecar's conflicting HIP_VISIBLE_DEVICES was neutralized
+    # (it would filter against the ROCR-pinned single-device view)
+    side_env = {e["name"]: e.get("value")
+                for e in mutated["spec"]["containers"][1]["env"]}
+    assert side_env["HIP_VISIBLE_DEVICES"] == "0"
+    assert side_env[C.ENV_ROCR_VISIBLE_DEVICES] == "2"
+    # volumes present exactly once
+    vols = [v["name"] for v in mutated["spec"]["volumes"]]
+    assert sorted(vols) == ["kubeshare-library", "kubeshare-sock"]
