@@ -412,3 +412,24 @@ def test_pod_mgr_uds_default_transport(tmp_path, native_bins, fake_gpus):
         assert _token_roundtrip(mgr_port, "ignored") > 0
     finally:
         nd.stop()
+
+
+def test_simulator_reference_trace_full():
+    """The reference's actual 989-job arrival trace
+    (test/simulator/trace.txt) replayed end to end against the
+    in-memory pipeline — the load test the reference could only run on
+    a live lab cluster. Skipped when the reference checkout is absent."""
+    import subprocess
+    import sys as _sys
+    trace = "/root/reference/test/simulator/trace.txt"
+    if not os.path.exists(trace):
+        pytest.skip("reference trace not available")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [_sys.executable, "tools/simulator.py", "--trace", trace,
+         "--nodes", "4"],
+        cwd=repo, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-1500:]
+    out = json.loads(r.stdout.strip().splitlines()[-1])
+    assert out["jobs"] == 989
+    assert out["p99_cycle_ms"] < 100.0
