@@ -31,6 +31,7 @@ class PodDemand:
     group_name: str = ""
     min_available: int = 0
     cell_id: str = ""
+    lease_ms: int = 0   # sharedgpu/lease_ms latency class
 
 
 def demand_from_pod(pod) -> PodDemand | None:
@@ -58,6 +59,7 @@ def demand_from_pod(pod) -> PodDemand | None:
         group_name=labels.get(C.POD_GROUP_NAME, ""),
         min_available=int(labels.get(C.POD_MIN_AVAILABLE, "0") or 0),
         cell_id=ann.get(C.POD_CELL_ID, ""),
+        lease_ms=int(labels.get(C.POD_LEASE_MS, "0") or 0),
     )
 
 

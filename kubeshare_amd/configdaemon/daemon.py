@@ -37,7 +37,8 @@ class ConfigDaemon:
             pod = f"{d.namespace}/{d.name}"
             by_uuid_cfg.setdefault(d.uuid, []).append(
                 F.PodQuota(pod, d.limit, d.request, d.memory,
-                           group=d.group_name))
+                           group=d.group_name,
+                           lease_ms=getattr(d, "lease_ms", 0)))
             by_uuid_port.setdefault(d.uuid, []).append(
                 F.PodPort(pod, d.port))
         if not by_uuid_cfg:

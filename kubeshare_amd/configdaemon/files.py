@@ -30,10 +30,15 @@ class PodQuota:
     request: float
     memory: int       # bytes; 0 = unlimited/default
     group: str = ""   # gang group: members are co-granted by gpu-schd
+    lease_ms: int = 0  # latency class: lease-length override (q=NNN)
 
     def line(self) -> str:
         base = f"{self.pod} {_fmt(self.limit)} {_fmt(self.request)} {self.memory}"
-        return base + (f" {self.group}\n" if self.group else "\n")
+        if self.group:
+            base += f" {self.group}"
+        if self.lease_ms:
+            base += f" q={self.lease_ms}"
+        return base + "\n"
 
 
 @dataclass(frozen=True)
@@ -83,9 +88,15 @@ def read_gpu_config(path: str) -> list[PodQuota]:
     for line in lines[1:1 + n]:
         parts = line.split()
         pod, limit, request, memory = parts[:4]
-        group = parts[4] if len(parts) > 4 else ""
+        group, lease = "", 0
+        for extra in parts[4:]:
+            if extra.startswith("q="):
+                lease = int(extra[2:])
+            else:
+                group = extra
         out.append(PodQuota(pod=pod, limit=float(limit), request=float(request),
-                            memory=int(memory), group=group))
+                            memory=int(memory), group=group,
+                            lease_ms=lease))
     return out
 
 

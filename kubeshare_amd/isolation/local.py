@@ -76,7 +76,8 @@ class PodHandle:
     request: float
     limit: float
     memory: int
-    manager_port: int
+    manager_port: int = 0
+    lease_ms: int = 0
     manager_uds: str = ""
     manager_proc: subprocess.Popen = None
 
@@ -138,12 +139,12 @@ class LocalGPUShare:
         return self
 
     def add_pod(self, name: str, request: float, limit: float | None = None,
-                memory: int = 0) -> PodHandle:
+                memory: int = 0, lease_ms: int = 0) -> PodHandle:
         limit = limit if limit is not None else 1.0
         if memory <= 0:
             memory = math.floor(request * self.full_memory)
         h = PodHandle(name=name, request=request, limit=limit, memory=memory,
-                      manager_port=0)
+                      lease_ms=lease_ms, manager_port=0)
         h.manager_uds = os.path.join(
             self.workdir, "pm-" + name.replace("/", "_") + ".sock")
         h.manager_port = free_port()  # TCP fallback listener
@@ -174,7 +175,8 @@ class LocalGPUShare:
     def _rewrite_config(self):
         F.write_gpu_config(
             self.config_dir, self.uuid,
-            [F.PodQuota(h.name, h.limit, h.request, h.memory)
+            [F.PodQuota(h.name, h.limit, h.request, h.memory,
+                        lease_ms=h.lease_ms)
              for h in self.pods.values()])
         F.write_port_config(
             self.port_dir, self.uuid,

@@ -24,6 +24,10 @@ POD_GPU_LIMIT = DOMAIN + "gpu_limit"
 POD_GPU_REQUEST = DOMAIN + "gpu_request"
 POD_GPU_MEMORY = DOMAIN + "gpu_mem"
 POD_GPU_MODEL = DOMAIN + "gpu_model"
+# optional latency class: per-pod lease-length override in ms (serving
+# pods want short leases; BASELINE.md serving table quantifies the
+# tradeoff). Clamped server-side to [min_quota, base_quota].
+POD_LEASE_MS = DOMAIN + "lease_ms"
 
 # annotations written by Reserve (reference pod.go:402-476)
 POD_GPU_UUID = DOMAIN + "gpu_uuid"

@@ -46,6 +46,7 @@ class PodSpec:
     pod_group: str = ""
     min_available: int = 0
     headcount: int = 0  # sharedgpu/group_headcount (Permit timeout unit)
+    lease_ms: int = 0   # sharedgpu/lease_ms latency class (0 = default)
     # filled at Reserve time
     uuids: list = field(default_factory=list)
     cell_ids: list = field(default_factory=list)
@@ -182,6 +183,17 @@ def parse_pod(namespace: str, name: str, labels: dict, *, uid: str = "",
             raise LabelError(f"{C.POD_GPU_MEMORY} set error: {raw_memory!r}") from e
         if memory < 0:
             raise LabelError(f"{C.POD_GPU_MEMORY} negative: {memory}")
+
+    raw_lease = labels.get(C.POD_LEASE_MS)
+    if raw_lease is not None:
+        try:
+            lease = int(raw_lease)
+        except ValueError as e:
+            raise LabelError(f"{C.POD_LEASE_MS} set error: {raw_lease!r}") from e
+        if not (0 < lease <= 10000):
+            raise LabelError(f"{C.POD_LEASE_MS} out of range (0,10000]: "
+                             f"{lease}")
+        spec.lease_ms = lease
 
     spec.limit = limit
     spec.request = request

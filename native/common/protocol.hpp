@@ -201,9 +201,12 @@ struct PodQuota {
   double limit = 1.0;
   double request = 0.0;
   long long memory = 0;
-  // optional 5th field: gang group — members are co-granted so a
-  // collective in one rank never spins on a token-starved peer
+  // optional trailing fields: gang group (bare word — members are
+  // co-granted so a collective in one rank never spins on a
+  // token-starved peer) and q=<ms> (latency class: per-pod
+  // lease-length override, clamped to [min_quota, base_quota])
   std::string group;
+  double lease_ms = 0.0;
 };
 
 inline bool parse_gpu_config(FILE* f, std::vector<PodQuota>& out) {
@@ -215,12 +218,19 @@ inline bool parse_gpu_config(FILE* f, std::vector<PodQuota>& out) {
     if (!fgets(line, sizeof(line), f)) return false;
     PodQuota q;
     char pod[2048];
-    char group[1024] = "";
-    int got = sscanf(line, "%2047s %lf %lf %lld %1023s", pod, &q.limit,
-                     &q.request, &q.memory, group);
+    char e5[1024] = "";
+    char e6[1024] = "";
+    int got = sscanf(line, "%2047s %lf %lf %lld %1023s %1023s", pod,
+                     &q.limit, &q.request, &q.memory, e5, e6);
     if (got < 4) return false;
     q.pod = pod;
-    if (got >= 5) q.group = group;
+    for (const char* extra : {e5, e6}) {
+      if (!extra[0]) continue;
+      if (extra[0] == 'q' && extra[1] == '=')
+        q.lease_ms = atof(extra + 2);
+      else
+        q.group = extra;
+    }
     out.push_back(q);
   }
   return true;

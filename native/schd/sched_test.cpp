@@ -415,6 +415,37 @@ int main() {
     CHECK(g.pod == "opp");
   }
 
+  // ---- 19. latency class: per-pod lease override (config q=<ms>)
+  {
+    TokenScheduler s(300, 10, 10000);
+    PodQuota q1 = quota("svc", 1.0, 0.3);
+    q1.lease_ms = 25.0;
+    s.set_config({q1, quota("train", 1.0, 0.7)});
+    Grant g;
+    double retry;
+    s.request("svc", 1, 200.0, 0.0);  // large hint must NOT win
+    CHECK(s.schedule(0.0, &g, &retry));
+    CHECK(g.pod == "svc" && g.quota_ms == 25.0);
+    s.release("svc", 25, 25.0);
+    s.request("train", 2, 0.0, 25.0);  // no override -> base quota
+    CHECK(s.schedule(25.0, &g, &retry));
+    CHECK(g.quota_ms == 300.0);
+    s.release("train", 300, 325.0);
+    // override above base clamps to base; below min clamps to min
+    PodQuota q2 = quota("hi", 1.0, 0.2);
+    q2.lease_ms = 900.0;
+    PodQuota q3 = quota("lo", 1.0, 0.2);
+    q3.lease_ms = 1.0;
+    s.set_config({q2, q3});
+    s.request("hi", 3, 0.0, 325.0);
+    CHECK(s.schedule(325.0, &g, &retry));
+    CHECK(g.quota_ms == 300.0);
+    s.release("hi", 10, 335.0);
+    s.request("lo", 4, 0.0, 335.0);
+    CHECK(s.schedule(335.0, &g, &retry));
+    CHECK(g.quota_ms == 10.0);  // min_q floor
+  }
+
   printf("sched_test OK (%d checks)\n", g_checks);
   return 0;
 }

@@ -4,7 +4,7 @@
 // Role model: the reference's gem-schd (one instance per physical GPU,
 // CLI flags -q 300 -m 20 -w 10000, SURVEY.md §2.2 and
 // docker/kubeshare-gemini-scheduler/launcher.py:27-31,77-80). Internals
-// are re-derived: deficit-based work-conserving time slicing.
+// are re-derived: stride-based work-conserving time slicing.
 //
 // Model
 // -----
@@ -56,6 +56,7 @@ struct PodAccount {
   double request = 0.0;
   double limit = 1.0;
   long long memory = 0;       // byte cap (0 = default)
+  double lease_ms = 0.0;      // latency class: lease override (0 = off)
   std::string group;          // gang group ("" = none)
   bool in_config = false;     // listed in the per-UUID config file?
   // (t_charge_end_ms, used_ms) entries inside the window
@@ -105,6 +106,7 @@ class TokenScheduler {
       a.request = q.request;
       a.limit = q.limit;
       a.memory = q.memory;
+      a.lease_ms = q.lease_ms;
       a.group = q.group;
       a.in_config = true;
     }
@@ -311,8 +313,11 @@ class TokenScheduler {
     // headroom lets a ramping pod grow back to base_q within a few
     // leases. Always clamped to [min_q, base_q].
     double desired = base_q_;
-    if (w.hint_ms > 0.0)
+    if (a.lease_ms > 0.0)
+      desired = a.lease_ms;  // explicit latency class beats the hint
+    else if (w.hint_ms > 0.0)
       desired = std::min(base_q_, std::max(min_q_, w.hint_ms * 1.5));
+    desired = std::min(desired, base_q_);
     double quota = std::min(desired, room);
     quota = std::max(quota, min_q_);
     holders_[w.pod] = Holder{now, quota};

@@ -63,3 +63,22 @@ def test_cpp_parses_python_writer(native_bins, tmp_path):
     finally:
         proc.kill()
         proc.wait()
+
+
+def test_lease_ms_roundtrip(tmp_path):
+    """Latency-class field: q=<ms> survives write -> python read and is
+    parsed by gpu-schd's C++ parser (covered via the loopback grant
+    test); order-independent with the gang group field."""
+    from kubeshare_amd.configdaemon import files as F
+    quotas = [
+        F.PodQuota("ns/svc", 1.0, 0.3, 0, lease_ms=25),
+        F.PodQuota("ns/gang", 1.0, 0.5, 123, group="g1", lease_ms=50),
+        F.PodQuota("ns/plain", 0.5, 0.25, 0),
+    ]
+    F.write_gpu_config(str(tmp_path), "GPU-q", quotas)
+    text = (tmp_path / "GPU-q").read_text()
+    assert " q=25" in text and "g1 q=50" in text
+    back = F.read_gpu_config(str(tmp_path / "GPU-q"))
+    assert back[0].lease_ms == 25 and back[0].group == ""
+    assert back[1].lease_ms == 50 and back[1].group == "g1"
+    assert back[2].lease_ms == 0

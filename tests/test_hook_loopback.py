@@ -153,3 +153,37 @@ def test_scheduler_ip_file_fallback(native_bins, loopback, tmp_path):
     finally:
         proc.kill()
         proc.wait()
+
+
+def test_latency_class_lease_override(native_bins, loopback, tmp_path):
+    """A pod with q=<ms> in its per-UUID config line gets leases of
+    that length (the sharedgpu/lease_ms latency class, end to end
+    through the real gpu-schd + hook)."""
+    cfg = tmp_path / "config"
+    cfg.mkdir(exist_ok=True)
+    (cfg / "GPU-x").write_text("1\nlo/svc 1.0 0.5 0 q=15\n")
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    proc = subprocess.Popen(
+        [native_bins["gpu-schd"], "-p", str(cfg), "-f", "GPU-x",
+         "-P", str(port), "-q", "60", "-m", "10", "-w", "3000"],
+        stderr=subprocess.DEVNULL)
+    try:
+        deadline = time.time() + 5
+        while time.time() < deadline:
+            try:
+                socket.create_connection(("127.0.0.1", port),
+                                         timeout=0.2).close()
+                break
+            except OSError:
+                time.sleep(0.05)
+        from kubeshare_amd.isolation.client import TokenClient
+        c = TokenClient("127.0.0.1", port, "lo/svc")
+        assert c.acquire(hint_ms=200.0) == 15.0  # override beats hint
+        c.release(15.0)
+        c.close()
+    finally:
+        proc.kill()
+        proc.wait()
