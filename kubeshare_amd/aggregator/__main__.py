@@ -6,13 +6,12 @@ from . import PodDemand, serve
 from ..utils import constants as C
 
 
-def kube_pod_source():
-    from kubernetes import client, config
-    try:
-        config.load_incluster_config()
-    except Exception:  # noqa: BLE001
-        config.load_kube_config()
-    v1 = client.CoreV1Api()
+def kube_pod_source(api=None):
+    if api is None:
+        from ..scheduler.kube import make_client
+        v1 = make_client()
+    else:
+        v1 = api
 
     def source():
         out = []
@@ -37,6 +36,7 @@ def kube_pod_source():
                     group_name=labels.get(C.POD_GROUP_NAME, ""),
                     min_available=int(labels.get(C.POD_MIN_AVAILABLE, 0)
                                       or 0),
+                    lease_ms=int(labels.get(C.POD_LEASE_MS, 0) or 0),
                     cell_id=ann.get(C.POD_CELL_ID, "")))
             except (ValueError, TypeError):
                 continue

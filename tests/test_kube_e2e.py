@@ -231,3 +231,30 @@ def test_multi_node_health_and_gang_over_http(tmp_path):
         assert pod.spec.node_name == "n2"
     finally:
         srv.stop()
+
+
+def test_lease_class_reaches_config_file(cluster, tmp_path):
+    """Latency-class chain over HTTP: sharedgpu/lease_ms label on a
+    Running pod -> kube_pod_source -> ConfigDaemon -> per-UUID config
+    line carries q=<ms> for gpu-schd."""
+    from kubeshare_amd.aggregator.__main__ import kube_pod_source
+    from kubeshare_amd.configdaemon import files as F
+    from kubeshare_amd.configdaemon.daemon import ConfigDaemon
+
+    srv, api, driver = cluster
+    srv.submit_pod("default", "svc",
+                   {C.POD_GPU_REQUEST: "0.3", C.POD_GPU_LIMIT: "1.0",
+                    C.POD_LEASE_MS: "25"})
+    driver.run_once()
+    source = kube_pod_source(api=api)
+    demands = [d for d in source() if d.node == "node-a"]
+    assert demands and demands[0].lease_ms == 25
+    cfg = tmp_path / "cfg"
+    prt = tmp_path / "prt"
+    cfg.mkdir()
+    prt.mkdir()
+    ConfigDaemon("node-a", str(cfg), str(prt)).update(demands)
+    uuid = demands[0].uuid
+    quotas = F.read_gpu_config(str(cfg / uuid))
+    assert quotas[0].lease_ms == 25
+    assert "q=25" in (cfg / uuid).read_text()
