@@ -102,23 +102,26 @@ _k8s_name = st.from_regex(r"[a-z0-9]([a-z0-9.-]{0,10}[a-z0-9])?",
 
 @settings(max_examples=60, deadline=None)
 @given(st.lists(
-    st.builds(lambda ns, n, lim, req, mem, grp: (f"{ns}/{n}", lim, req,
-                                                 mem, grp),
+    st.builds(lambda ns, n, lim, req, mem, grp, lease: (f"{ns}/{n}", lim,
+                                                        req, mem, grp,
+                                                        lease),
               _k8s_name, _k8s_name,
               st.floats(min_value=0.01, max_value=1.0),
               st.floats(min_value=0.0, max_value=1.0),
               st.integers(min_value=0, max_value=2**48),
-              st.one_of(st.just(""), _k8s_name)),
+              st.one_of(st.just(""), _k8s_name),
+              st.integers(min_value=0, max_value=10000)),
     min_size=0, max_size=8))
 def test_config_file_roundtrip_property(tmp_path_factory, entries):
     """Arbitrary (k8s-legal) pods/groups/values survive the per-UUID
     file write/read round trip bit-exactly enough for scheduling
-    (request/limit to 1e-6, memory and group exactly)."""
+    (request/limit to 1e-6, memory/group/lease exactly)."""
     from kubeshare_amd.configdaemon import files as F
     tmp = tmp_path_factory.mktemp("cfgprop")
-    # PodQuota(pod, limit, request, memory, group); request <= limit
-    quotas = [F.PodQuota(pod, lim, min(req, lim), mem, group=grp)
-              for pod, lim, req, mem, grp in entries]
+    # PodQuota(pod, limit, request, memory, group, lease); request<=limit
+    quotas = [F.PodQuota(pod, lim, min(req, lim), mem, group=grp,
+                         lease_ms=lease)
+              for pod, lim, req, mem, grp, lease in entries]
     path = F.write_gpu_config(str(tmp), "GPU-prop", quotas)
     back = F.read_gpu_config(path)
     assert len(back) == len(quotas)
@@ -128,3 +131,4 @@ def test_config_file_roundtrip_property(tmp_path_factory, entries):
         assert abs(a.request - b.request) < 1e-6
         assert a.memory == b.memory
         assert a.group == b.group
+        assert a.lease_ms == b.lease_ms
