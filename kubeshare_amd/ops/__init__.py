@@ -71,10 +71,17 @@ class FusedSGD:
     Matches torch.optim.SGD(momentum=mu, weight_decay=wd, dampening=0,
     nesterov=False) semantics; one kernel per tensor, no Python-side
     per-parameter loop allocations after the first step.
+
+    graph_mode="auto" (default on CUDA): once the grad set is stable,
+    the ~161 per-tensor update launches and the ~161 grad-zero memsets
+    are captured into two hipGraphs and replayed as ONE gated dispatch
+    each per step. Requires stable grad storage, so zero_grad() zeroes
+    in place instead of dropping the tensors (backward then accumulates
+    into the same buffers — pointer-stable, capture-safe).
     """
 
     def __init__(self, params, lr: float, momentum: float = 0.9,
-                 weight_decay: float = 0.0):
+                 weight_decay: float = 0.0, graph_mode: str = "auto"):
         import torch
 
         self.params = [p for p in params if p.requires_grad]
@@ -85,17 +92,63 @@ class FusedSGD:
         # conv weights included): the kernel updates flat dense storage
         self.momenta = [torch.zeros_like(p) for p in self.params]
         self._first = True
+        self._graph_mode = graph_mode
+        self._step_graph = None
+        self._zero_graph = None
+        self._graph_ptrs = None  # grad data_ptrs the graphs were baked on
+
+    def _graphable(self):
+        import torch
+        if self._graph_mode == "off" or not torch.cuda.is_available():
+            return False
+        import os
+        if os.environ.get("KUBESHARE_SGD_GRAPH", "1") == "0":
+            return False
+        return all(p.grad is not None and p.grad.is_cuda
+                   for p in self.params)
 
     def zero_grad(self, set_to_none: bool = True):
+        if self._zero_graph is not None and self._graph_ptrs == tuple(
+                p.grad.data_ptr() if p.grad is not None else 0
+                for p in self.params):
+            self._zero_graph.replay()
+            return
         for p in self.params:
             if p.grad is not None:
-                if set_to_none:
+                if set_to_none and self._step_graph is None:
                     p.grad = None
                 else:
+                    # graph mode needs pointer-stable grads
                     p.grad.zero_()
+
+    def _capture(self):
+        import torch
+        ext = _load()
+        ps = [p.data for p in self.params]
+        gs = [p.grad for p in self.params]
+        vs = list(self.momenta)
+        ptrs = tuple(g.data_ptr() for g in gs)
+        # (the update kernel is already warm: capture happens right
+        # after a full eager step)
+        step_g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(step_g):
+            ext.sgd_momentum_(ps, gs, vs, self.lr, self.momentum,
+                              self.weight_decay)
+        zero_g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(zero_g):
+            for g in gs:
+                g.zero_()
+        self._step_graph = step_g
+        self._zero_graph = zero_g
+        self._graph_ptrs = ptrs
 
     def step(self):
         ext = _load()
+        if self._step_graph is not None and self._graph_ptrs == tuple(
+                p.grad.data_ptr() if p.grad is not None else 0
+                for p in self.params):
+            self._step_graph.replay()
+            return
         ps, gs, vs = [], [], []
         for p, v in zip(self.params, self.momenta):
             if p.grad is None:
@@ -109,6 +162,14 @@ class FusedSGD:
             self._first = False
         ext.sgd_momentum_(ps, gs, vs, self.lr, self.momentum,
                           self.weight_decay)
+        # after a full-set eager step with stable grads, capture the
+        # graphs for subsequent steps
+        if self._step_graph is None and self._graph_mode == "auto" and \
+                len(gs) == len(self.params) and self._graphable():
+            try:
+                self._capture()
+            except Exception:  # noqa: BLE001 — capture is an optimization
+                self._graph_mode = "off"
 
 
 # ---------------------------------------------------- fused BN+ReLU(+add)

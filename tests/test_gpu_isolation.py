@@ -96,6 +96,17 @@ def test_sgd_momentum_matches_torch():
     torch.cuda.synchronize()
     for p, r in zip(ps, ref_ps):
         torch.testing.assert_close(p, r, rtol=1e-5, atol=1e-6)
+    # steps 2-3 ran as hipGraph replays (capture after the first full
+    # eager step); a FRESH grad tensor set (new storage -> pointer
+    # mismatch) must fall back to the eager path and still match torch
+    assert fused._step_graph is not None, "graph capture never engaged"
+    for p, g in zip(ps, gs):
+        p.grad = g.clone()
+    fused.step()
+    opt.step()  # torch continues with its persistent grads/momentum
+    torch.cuda.synchronize()
+    for p, r in zip(ps, ref_ps):
+        torch.testing.assert_close(p, r, rtol=1e-5, atol=1e-6)
 
 
 def test_two_pods_5050_split(share):
