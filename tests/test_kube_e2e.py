@@ -258,3 +258,39 @@ def test_lease_class_reaches_config_file(cluster, tmp_path):
     quotas = F.read_gpu_config(str(cfg / uuid))
     assert quotas[0].lease_ms == 25
     assert "q=25" in (cfg / uuid).read_text()
+
+
+def test_restart_with_waiting_gang_leaks_nothing(cluster):
+    """Driver crash while a partial gang is parked at Permit: the new
+    driver (fresh process) rebuilds from BOUND pods only — the waiting
+    members were never applied, so a clean resync shows no leaked
+    reservations and the gang can complete later."""
+    from kubeshare_amd.scheduler.kube import KubeDriver
+    from kubeshare_amd.scheduler.topology import TopologyConfig
+    srv, api, driver = cluster
+    labels = {C.POD_GPU_REQUEST: "0.5", C.POD_GPU_LIMIT: "1.0",
+              C.POD_GROUP_NAME: "g2", C.POD_GROUP_HEADCOUNT: "2",
+              C.POD_GROUP_THRESHOLD: "1.0"}
+    srv.submit_pod("default", "g2-a", labels)
+    # second member exists but is not schedulable this cycle
+    pod_b = srv.submit_pod("default", "g2-b", dict(labels))
+    with srv.lock:
+        pod_b["status"]["phase"] = "Unknown"
+    driver.run_once()
+    assert "default/g2-a" in driver.waiting_pods   # parked, not applied
+
+    # "crash": a fresh driver resyncs from the cluster
+    fresh = KubeDriver(TopologyConfig.single_node("node-a", gpus=2),
+                       inventory=FakeInventory({"node-a": {"gpus": 2}}),
+                       api=api)
+    fresh.sync_nodes()
+    fresh.resync_bound()
+    for leaf in fresh.sched.tree.leaves_on_node("node-a"):
+        assert leaf.available == 1.0   # nothing leaked into the resync
+    # the gang completes once both members are Pending
+    with srv.lock:
+        pod_b["status"]["phase"] = "Pending"
+    fresh.run_once()
+    for name in ("g2-a", "g2-b"):
+        pod = api.read_namespaced_pod(name, "default")
+        assert pod.spec.node_name == "node-a", name
