@@ -80,10 +80,15 @@ def format_node_annotation(gpus: list) -> str:
     indices with a live direct xGMI link."""
     parts = []
     for g in gpus:
-        links = ":".join(str(j) for j, w in sorted(g.xgmi_links.items())
-                         if w > 0)
-        parts.append(f"{g.uuid},{g.model},{g.memory},{g.index},"
-                     f"links={links}")
+        entry = f"{g.uuid},{g.model},{g.memory},{g.index}"
+        if g.xgmi_links:
+            links = ":".join(str(j) for j, w in sorted(g.xgmi_links.items())
+                             if w > 0)
+            entry += f",links={links}"
+        # no link info -> omit the field entirely: an EMPTY links= would
+        # read back as "every link down" instead of "unknown topology,
+        # assume the MI355X clique"
+        parts.append(entry)
     return ";".join(parts)
 
 

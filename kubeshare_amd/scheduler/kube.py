@@ -83,7 +83,8 @@ class KubeDriver:
                     gpu = {"uuid": uuid, "memory": int(memory),
                            "index": int(index)}
                     for extra in fields[4:]:
-                        if extra.startswith("links="):
+                        if extra.startswith("links=") and \
+                                extra != "links=":
                             gpu["xgmi_links"] = {
                                 int(j): 1
                                 for j in extra[len("links="):].split(":")

@@ -282,3 +282,29 @@ def test_small_workload_families_train_on_cpu():
             opt.step()
             l0 = l0 if l0 is not None else loss.item()
         assert loss.item() < l0, name  # it actually learns the batch
+
+
+def test_annotation_without_links_assumes_clique():
+    """A GPU with no link info must serialize WITHOUT the links field:
+    an empty links= would read back as 'every link down' instead of
+    'unknown topology -> clique'."""
+    from kubeshare_amd.scheduler.inventory import (GPUInfo,
+                                                   format_node_annotation)
+    g = GPUInfo(uuid="GPU-n-0", model=C.MI355X_MODEL,
+                memory=C.MI355X_HBM_BYTES, index=0)
+    ann = format_node_annotation([g])
+    assert "links" not in ann
+    # and the parse side leaves xgmi_links unset -> clique assumption
+    from types import SimpleNamespace as NS
+    from kubeshare_amd.scheduler.kube import KubeDriver
+
+    class V1:
+        def list_node(self, label_selector=None):
+            return NS(items=[NS(
+                metadata=NS(name="n", annotations={
+                    "kubeshare.amd/gpus": ann + ",links="}),
+                status=NS(conditions=[NS(type="Ready", status="True")]))])
+    d = KubeDriver(TopologyConfig.single_node("n", gpus=1), api=V1())
+    d.sync_nodes()
+    leaf = d.sched.tree.leaf_by_uuid["GPU-n-0"]
+    assert leaf.xgmi_peers is None  # unknown -> clique in _distance
