@@ -48,7 +48,9 @@ class GraphReplayServer:
                 self.static_out = self.model(self.static_in)
 
     def serve(self, x) -> "object":
-        """One request: copy in, replay, sync, return output (cloned)."""
+        """One request: copy in, replay, sync. Returns the STATIC
+        output buffer — the next request overwrites it; callers that
+        keep results across requests must .clone()."""
         self.static_in.copy_(x)
         self.graph.replay()
         self.torch.cuda.synchronize()
