@@ -74,13 +74,14 @@ class GPURequirementCollector:
             "GPU demand of running shared-GPU pods",
             labels=["namespace", "pod", "pod_id", "node", "group_name",
                     "min_available", "limit", "request", "memory",
-                    "cell_id", "uuid", "port"])
+                    "cell_id", "uuid", "port", "lease_ms"])
         now = time.time()
         for d in self.pod_source():
             fam.add_metric(
                 [d.namespace, d.name, d.pod_id, d.node, d.group_name,
                  str(d.min_available), str(d.limit), str(d.request),
-                 str(d.memory), d.cell_id, d.uuid, str(d.port)], now)
+                 str(d.memory), d.cell_id, d.uuid, str(d.port),
+                 str(d.lease_ms)], now)
         yield fam
 
 

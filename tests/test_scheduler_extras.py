@@ -308,3 +308,51 @@ def test_annotation_without_links_assumes_clique():
     d.sync_nodes()
     leaf = d.sched.tree.leaf_by_uuid["GPU-n-0"]
     assert leaf.xgmi_peers is None  # unknown -> clique in _distance
+
+
+def test_two_level_topology_gang_prefers_one_node():
+    """2-level cell tree (the shipped deploy/config example): a gang of
+    whole-GPU pods that FITS on one node stays there (intra-node xGMI
+    distance 1 < cross-node digit distance), and a gang larger than a
+    node spills to the sibling node under the same parent cell."""
+    from kubeshare_amd.scheduler.harness import FakeCluster
+    from kubeshare_amd.scheduler.inventory import FakeInventory
+    from kubeshare_amd.scheduler.topology import TopologyConfig
+    topo = TopologyConfig.from_yaml("""
+cellTypes:
+  MI355X-NODE:
+    childCellType: "AMD Instinct MI355X"
+    childCellNumber: 2
+    childCellPriority: 100
+    isNodeLevel: true
+  2-MI355X-NODE:
+    childCellType: MI355X-NODE
+    childCellNumber: 2
+cells:
+- cellType: 2-MI355X-NODE
+  cellChildren:
+  - cellId: node-a
+  - cellId: node-b
+""")
+    fc = FakeCluster(topology=topo, nodes={"node-a": {"gpus": 2},
+                                           "node-b": {"gpus": 2}})
+    inv = FakeInventory({"node-a": {"gpus": 2}, "node-b": {"gpus": 2}})
+    fc.inventory = inv
+    for n in ("node-a", "node-b"):
+        fc.scheduler.register_node(n, inv.by_model(n))
+    labels = {C.POD_GPU_REQUEST: "1.0", C.POD_GPU_LIMIT: "1.0",
+              C.POD_PRIORITY: "100", C.POD_GROUP_NAME: "g",
+              C.POD_GROUP_HEADCOUNT: "2", C.POD_GROUP_THRESHOLD: "1.0"}
+    fc.add_pod("ns", "r0", labels)
+    fc.add_pod("ns", "r1", labels)
+    fc.schedule_pending()
+    nodes = {fc.pods[f"ns/r{i}"].node for i in (0, 1)}
+    assert len(nodes) == 1, nodes  # 2-GPU gang packs onto one node
+
+    # a second 2-GPU gang fills the OTHER node (first is full)
+    labels2 = dict(labels, **{C.POD_GROUP_NAME: "h"})
+    fc.add_pod("ns", "s0", labels2)
+    fc.add_pod("ns", "s1", labels2)
+    fc.schedule_pending()
+    nodes2 = {fc.pods[f"ns/s{i}"].node for i in (0, 1)}
+    assert len(nodes2) == 1 and nodes2 != nodes
