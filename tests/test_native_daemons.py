@@ -469,3 +469,29 @@ def test_sched_unit(native_bins):
                        text=True, timeout=60)
     assert r.returncode == 0, r.stderr or r.stdout
     assert "sched_test OK" in r.stdout
+
+
+def test_lease_class_hot_reload(native_bins, tmp_path):
+    """Adding/removing q=<ms> in the per-UUID file mid-run re-sizes the
+    pod's leases on the next grant (inotify reload, no restart)."""
+    from kubeshare_amd.isolation.client import TokenClient
+    cfg = _write_config(tmp_path, [("ns/svc", 1.0, 0.5, 0)])
+    schd = Schd(native_bins, cfg)
+    try:
+        c = TokenClient("127.0.0.1", schd.port, "ns/svc")
+        assert c.acquire() == 60.0           # base quota (-q 60)
+        c.release(5.0)
+        (tmp_path / "config" / "GPU-x").write_text(
+            "1\nns/svc 1.0 0.5 0 q=15\n")
+        time.sleep(0.5)                      # inotify turnaround
+        assert c.acquire() == 15.0           # latency class applied
+        c.release(5.0)
+        (tmp_path / "config" / "GPU-x").write_text(
+            "1\nns/svc 1.0 0.5 0\n")
+        time.sleep(0.5)
+        q = c.acquire()
+        c.release(5.0)
+        c.close()
+        assert q == 60.0                     # override removed
+    finally:
+        schd.stop()
