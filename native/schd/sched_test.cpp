@@ -446,6 +446,27 @@ int main() {
     CHECK(g.quota_ms == 10.0);  // min_q floor
   }
 
+  // ---- 20. config-line parser: optional trailing fields in any order
+  {
+    FILE* f = tmpfile();
+    fputs("4\n"
+          "ns/a 1.0 0.5 0\n"
+          "ns/b 1.0 0.5 123 gangx\n"
+          "ns/c 1.0 0.5 0 q=25\n"
+          "ns/d 0.9 0.4 7 q=15 gangy\n",  // swapped order also legal
+          f);
+    rewind(f);
+    std::vector<PodQuota> out;
+    CHECK(parse_gpu_config(f, out));
+    fclose(f);
+    CHECK(out.size() == 4);
+    CHECK(out[0].group.empty() && out[0].lease_ms == 0.0);
+    CHECK(out[1].group == "gangx" && out[1].lease_ms == 0.0);
+    CHECK(out[2].group.empty() && out[2].lease_ms == 25.0);
+    CHECK(out[3].group == "gangy" && out[3].lease_ms == 15.0);
+    CHECK(out[3].memory == 7 && out[3].request == 0.4);
+  }
+
   printf("sched_test OK (%d checks)\n", g_checks);
   return 0;
 }
