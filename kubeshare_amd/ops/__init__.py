@@ -184,7 +184,9 @@ def _bn_relu_autograd():
         """y = relu(batch_norm(x) [+ res]) on NHWC bf16 via the gfx950
         kernels in hip/bn_relu.hip. Replaces MIOpen's 5-pass forward /
         8-pass backward (incl. separate ReLU and residual-add kernels)
-        with 3 / 7 passes — see profiles/README.md for the motivation."""
+        with 3 forward passes and 5 (non-residual, mask recomputed
+        from x) / 7 (residual, dym reuse) backward passes — see
+        profiles/README.md for the motivation."""
 
         @staticmethod
         def forward(ctx, x, res, weight, bias, running_mean, running_var,
