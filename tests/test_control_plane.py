@@ -272,6 +272,11 @@ def test_noded_metrics_exporter(tmp_path, native_bins, fake_gpus):
         assert 'gpu_pod_token_grants_total{node="node-a",pod="ns/m1",' \
                'uuid="GPU-fake-0"}' in text
         assert "gpu_pod_window_usage_ms" in text
+        # round-2 accounting observability: sampler mode (0 on CPU
+        # boxes — wall/RET fallback) and unattributed busy
+        assert 'gpu_schd_busy_sampler{node="node-a",' \
+               'uuid="GPU-fake-0"} 0.0' in text
+        assert "gpu_schd_unattributed_busy_ms" in text
     finally:
         nd.stop()
 
