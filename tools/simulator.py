@@ -46,16 +46,31 @@ def load_trace(path: str):
     return out
 
 
-def job_labels(rng, gpus: int):
-    """Randomized fractional requests as the reference simulator does
-    (simulator.py:64-69)."""
+def job_pods(rng, job_id: int, gpus: int):
+    """Label sets for one trace job. Multi-GPU jobs are either one
+    N-GPU pod or (half the time) a GANG of N whole-GPU pods — the
+    reference's distributed e2e shape (test/distribute, one NCCL rank
+    per GPU) — so the Permit barrier/expiry machinery is soaked too.
+    Fractional jobs randomize request and priority like the reference
+    simulator (simulator.py:64-69); some carry a lease_ms latency
+    class."""
+    prio = str(rng.choice([0, 50, 100]))
+    if gpus > 1 and rng.random() < 0.5:
+        gang = {C.POD_GPU_REQUEST: "1.0", C.POD_GPU_LIMIT: "1.0",
+                C.POD_PRIORITY: prio,
+                C.POD_GROUP_NAME: f"sim-gang-{job_id}",
+                C.POD_GROUP_HEADCOUNT: str(gpus),
+                C.POD_GROUP_THRESHOLD: "1.0"}
+        return [dict(gang) for _ in range(gpus)]
     if gpus > 1:
-        return {C.POD_GPU_REQUEST: f"{gpus}.0", C.POD_GPU_LIMIT: f"{gpus}.0",
-                C.POD_PRIORITY: str(rng.choice([0, 50, 100]))}
-    request = rng.choice([0.25, 0.5, 0.75, 1.0])
-    limit = 1.0
-    return {C.POD_GPU_REQUEST: str(request), C.POD_GPU_LIMIT: str(limit),
-            C.POD_PRIORITY: str(rng.choice([0, 0, 100]))}
+        return [{C.POD_GPU_REQUEST: f"{gpus}.0",
+                 C.POD_GPU_LIMIT: f"{gpus}.0", C.POD_PRIORITY: prio}]
+    labels = {C.POD_GPU_REQUEST: str(rng.choice([0.25, 0.5, 0.75, 1.0])),
+              C.POD_GPU_LIMIT: "1.0",
+              C.POD_PRIORITY: str(rng.choice([0, 0, 100]))}
+    if rng.random() < 0.2:
+        labels[C.POD_LEASE_MS] = str(rng.choice([25, 50, 100]))
+    return [labels]
 
 
 def main():
@@ -75,54 +90,71 @@ def main():
     fc = FakeCluster(nodes={f"node-{i}": {"gpus": args.gpus_per_node}
                             for i in range(args.nodes)})
 
-    active = []   # (end_time, key)
-    stats = {"bound": 0, "unschedulable": 0, "retries": 0,
-             "cycle_ms": []}
-    for i, (start, gpus, runtime) in enumerate(trace):
-        fc.clock = start
-        # reap finished jobs
-        for end, key in [a for a in active if a[0] <= start]:
+    runtime_of = {}        # pod key -> job runtime (s)
+    active = []            # (end_time, key) of bound pods
+    in_active = set()
+    cycles = []
+    requeues = 0
+    ev_seen = 0
+
+    def absorb_bindings():
+        """Move newly-bound pods (harness event log) into the active
+        set with their job runtime."""
+        nonlocal ev_seen
+        for e in fc.events[ev_seen:]:
+            if e[0] == "bind" and e[1] not in in_active:
+                active.append((fc.clock + runtime_of.get(e[1], 120.0),
+                               e[1]))
+                in_active.add(e[1])
+        ev_seen = len(fc.events)
+
+    def reap(now):
+        for end, key in [a for a in active if a[0] <= now]:
             fc.delete_pod(key)
             active.remove((end, key))
-        pod = fc.add_pod("sim", f"job{i}", job_labels(rng, gpus))
+            in_active.discard(key)
+
+    for i, (start_t, gpus, runtime) in enumerate(trace):
+        fc.clock = start_t
+        reap(start_t)
+        for k, labels in enumerate(job_pods(rng, i, gpus)):
+            pod = fc.add_pod("sim", f"job{i}-{k}", labels)
+            runtime_of[pod.key] = runtime
         t0 = time.perf_counter()
         fc.schedule_pending(rounds=1)
-        stats["cycle_ms"].append((time.perf_counter() - t0) * 1000)
-        if pod.phase == "Bound":
-            stats["bound"] += 1
-            active.append((start + runtime, pod.key))
-        elif pod.phase == "Unschedulable":
-            # retry while capacity frees up (the real queue requeues)
-            stats["retries"] += 1
-            pod.phase = "Pending"
+        cycles.append((time.perf_counter() - t0) * 1000)
+        absorb_bindings()
+        # Unschedulable (incl. Permit-expired gang members) requeues —
+        # the real scheduler's backoff queue
+        for pod in fc.pods.values():
+            if pod.phase == "Unschedulable":
+                requeues += 1
+                pod.phase = "Pending"
 
-    # drain: retry pending as jobs finish
-    pending = [p for p in fc.pods.values() if p.phase == "Pending"]
-    t = fc.clock
-    for _ in range(200):
+    # drain: keep reaping finished jobs and rescheduling until quiet
+    for _ in range(500):
+        pending = [p for p in fc.pods.values()
+                   if p.phase in ("Pending", "Waiting")]
         if not pending or not active:
             break
         active.sort()
-        t, key = active.pop(0)
-        fc.delete_pod(key)
-        fc.clock = t
+        fc.clock = active[0][0]
+        reap(fc.clock)
         fc.schedule_pending(rounds=1)
-        newly = [p for p in pending if p.phase == "Bound"]
-        for p in newly:
-            stats["bound"] += 1
-            active.append((t + 120, p.key))
-        pending = [p for p in pending if p.phase == "Pending"]
+        absorb_bindings()
+        for pod in fc.pods.values():
+            if pod.phase == "Unschedulable":
+                pod.phase = "Pending"
 
-    cycles = stats.pop("cycle_ms")
-    # every bind is recorded in the harness event log, including jobs
-    # that bound during a LATER arrival's cycle after a first-try miss
     ever_bound = {e[1] for e in fc.events if e[0] == "bind"}
+    never = [p for p in fc.pods.values()
+             if p.phase in ("Pending", "Waiting")]
     result = {
         "jobs": len(trace),
-        "ever_bound": len(ever_bound),
-        "bound_first_try": stats["bound"],
-        "unschedulable_final": len(pending),
-        "first_try_retries": stats["retries"],
+        "pods_submitted": len(runtime_of),
+        "pods_ever_bound": len(ever_bound),
+        "never_bound_after_drain": len(never),
+        "requeues": requeues,
         "mean_cycle_ms": round(sum(cycles) / len(cycles), 3),
         "p99_cycle_ms": round(sorted(cycles)[int(len(cycles) * 0.99)], 3),
     }
