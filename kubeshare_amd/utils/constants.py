@@ -24,9 +24,12 @@ POD_GPU_LIMIT = DOMAIN + "gpu_limit"
 POD_GPU_REQUEST = DOMAIN + "gpu_request"
 POD_GPU_MEMORY = DOMAIN + "gpu_mem"
 POD_GPU_MODEL = DOMAIN + "gpu_model"
-# optional latency class: per-pod lease-length override in ms (serving
-# pods want short leases; BASELINE.md serving table quantifies the
-# tradeoff). Clamped server-side to [min_quota, base_quota].
+# optional lease-length bound in ms: caps how long THIS pod's leases
+# can block co-located pods (a request arriving mid-lease waits for the
+# holder's drain). Set a small value on throughput pods sharing a GPU
+# with latency-critical serving — the serving pod's own leases are
+# already right-sized by its busy-EWMA hint. Clamped server-side to
+# [min_quota, base_quota]; BASELINE.md quantifies the tradeoff.
 POD_LEASE_MS = DOMAIN + "lease_ms"
 
 # annotations written by Reserve (reference pod.go:402-476)
