@@ -132,3 +132,49 @@ def test_config_file_roundtrip_property(tmp_path_factory, entries):
         assert a.memory == b.memory
         assert a.group == b.group
         assert a.lease_ms == b.lease_ms
+
+
+_env_name = st.from_regex(r"[A-Z][A-Z0-9_]{0,12}", fullmatch=True)
+_container = st.builds(
+    lambda name, env, mounts: {
+        "name": name,
+        **({"env": env} if env is not None else {}),
+        **({"volumeMounts": mounts} if mounts is not None else {})},
+    _k8s_name,
+    st.one_of(st.none(), st.lists(st.builds(
+        lambda n, v: {"name": n, "value": v}, _env_name, _k8s_name),
+        max_size=4)),
+    st.one_of(st.none(), st.lists(st.builds(
+        lambda n: {"name": n, "mountPath": "/" + n}, _k8s_name),
+        max_size=2)))
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.lists(_container, min_size=1, max_size=3),
+       st.sampled_from(["50050", "50300", ""]),
+       st.one_of(st.none(), st.lists(st.builds(
+           lambda n: {"name": n}, _k8s_name), max_size=2)))
+def test_webhook_patch_idempotent_property(containers, port, volumes):
+    """For ANY container/env/volume shape: the webhook patch applies
+    cleanly (RFC-6902 paths valid against the doc) and a second
+    invocation is a no-op."""
+    from kubeshare_amd.testing.fake_apiserver import apply_json_patch
+    from kubeshare_amd.webhook import build_patch
+    pod = {"metadata": {"name": "p", "namespace": "ns",
+                        "annotations": {C.POD_GPU_UUID: "GPU-1",
+                                        C.POD_GPU_INDEX: "0",
+                                        C.POD_GPU_MEMORY: "5",
+                                        **({C.POD_MANAGER_PORT: port}
+                                           if port else {})}},
+           "spec": {"containers": containers,
+                    **({"volumes": volumes} if volumes is not None
+                       else {})}}
+    patch = build_patch(pod)
+    mutated = apply_json_patch(pod, patch)     # must not raise
+    assert build_patch(mutated) == []          # idempotent
+    for c in mutated["spec"]["containers"]:
+        names = [e["name"] for e in c["env"]]
+        assert names.count(C.ENV_ROCR_VISIBLE_DEVICES) == 1
+        assert C.ENV_INJECTED in names
+        if port:
+            assert names.count("POD_MANAGER_UDS") == 1
