@@ -169,6 +169,14 @@ class KubeDriver:
         if spec is None:
             if err:
                 self.event(pod, "PreFilterRejected", err)
+            else:
+                # REGULAR pod (no sharedgpu labels) that still named
+                # this scheduler: inside kube-scheduler the framework's
+                # default plugins would place it (reference
+                # cmd/kubeshare-scheduler/main.go:26-38); out-of-tree we
+                # must not strand it Pending — bind to the healthy node
+                # with the fewest pods, no env injection.
+                self.bind_regular(pod)
             return
         feasible = []
         for node in self.sched.tree.all_nodes():
@@ -195,9 +203,28 @@ class KubeDriver:
         else:
             self.waiting_pods[spec.key] = (pod, placement)
 
+    def bind_regular(self, pod):
+        """Least-loaded-node placement for label-less pods that chose
+        this scheduler anyway."""
+        healthy = [n for n in self.sched.tree.all_nodes()
+                   if any(c.healthy for c in self.sched.tree.node_cells[n])]
+        if not healthy:
+            self.event(pod, "Unschedulable", "no healthy node")
+            return
+        load = {n: 0 for n in healthy}
+        for p in self.v1.list_pod_for_all_namespaces().items:
+            if p.spec.node_name in load:
+                load[p.spec.node_name] += 1
+        node = min(sorted(healthy), key=lambda n: load[n])
+        self.apply_placement(pod, Placement(
+            node=node, uuids=[], cell_ids=[], gpu_indices=[],
+            gpu_mem=0, manager_port=0))
+
     def placement_env(self, pod, placement: Placement) -> list:
         """Env entries (plain dicts — both clients serialize them) for
         the injected containers; identical to the webhook path."""
+        if not placement.uuids:
+            return []  # regular pod: no injection at all
         idx = ",".join(map(str, placement.gpu_indices))
         if placement.manager_port:
             return shared_pod_env(pod.metadata.namespace, pod.metadata.name,

@@ -294,3 +294,21 @@ def test_restart_with_waiting_gang_leaks_nothing(cluster):
     for name in ("g2-a", "g2-b"):
         pod = api.read_namespaced_pod(name, "default")
         assert pod.spec.node_name == "node-a", name
+
+
+def test_regular_pod_not_stranded(cluster):
+    """A pod that names kubeshare-scheduler but has NO sharedgpu labels
+    (the reference's 'regular pod', pod.go:303-305) must still get
+    bound — inside kube-scheduler the default plugins would place it;
+    the out-of-tree driver binds it least-loaded with no injection."""
+    srv, api, driver = cluster
+    srv.submit_pod("default", "plain", {})      # no labels at all
+    driver.run_once()
+    pod = api.read_namespaced_pod("plain", "default")
+    assert pod.spec.node_name == "node-a"
+    assert pod.status.phase == "Running"
+    env = pod.spec.containers[0].to_payload().get("env")
+    assert not env                              # no injection
+    # and no scheduler-side GPU reservation was charged
+    for leaf in driver.sched.tree.leaves_on_node("node-a"):
+        assert leaf.available == 1.0
