@@ -195,13 +195,25 @@ class KubeDriver:
             return
         decision, _, release = self.sched.permit(spec)
         if decision == "allow":
-            self.apply_placement(pod, placement)
+            self._apply_or_unreserve(pod, placement, spec.key)
             for key in release:
                 w = self.waiting_pods.pop(key, None)
                 if w is not None:
-                    self.apply_placement(w[0], w[1])
+                    self._apply_or_unreserve(w[0], w[1], key)
         else:
             self.waiting_pods[spec.key] = (pod, placement)
+
+    def _apply_or_unreserve(self, pod, placement, key):
+        """Apply a placement; if the pod vanished (deleted mid-release)
+        roll the reservation back so a later retry cannot double-charge
+        the cells/port."""
+        try:
+            self.apply_placement(pod, placement)
+        except Exception as e:  # noqa: BLE001
+            self.event(pod, "ApplyFailed", str(e))
+            spec = self.sched.pod_status.get(key)
+            if spec is not None:
+                self.sched.unreserve(spec)
 
     def bind_regular(self, pod):
         """Least-loaded-node placement for label-less pods that chose

@@ -312,3 +312,21 @@ def test_regular_pod_not_stranded(cluster):
     # and no scheduler-side GPU reservation was charged
     for leaf in driver.sched.tree.leaves_on_node("node-a"):
         assert leaf.available == 1.0
+
+
+def test_apply_failure_rolls_back_reservation(cluster):
+    """If a pod is deleted between Reserve and apply (shadow-pod
+    recreate), the driver must unreserve — otherwise the next pod sees
+    phantom usage and a retried namesake double-charges."""
+    srv, api, driver = cluster
+    srv.submit_pod("default", "ghost",
+                   {C.POD_GPU_REQUEST: "0.5", C.POD_GPU_LIMIT: "1.0"})
+    pods = driver.v1.list_pod_for_all_namespaces(
+        field_selector="status.phase=Pending").items
+    target = [p for p in pods if p.metadata.name == "ghost"][0]
+    with srv.lock:
+        srv.pods.pop(("default", "ghost"))   # user deletes it racily
+    driver.schedule_pod(target)              # apply must fail + roll back
+    for leaf in driver.sched.tree.leaves_on_node("node-a"):
+        assert leaf.available == 1.0, leaf
+    assert driver.sched.ports["node-a"].available()
