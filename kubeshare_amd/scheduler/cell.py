@@ -120,8 +120,8 @@ class CellTree:
             # index-keyed xGMI adjacency into uuid-keyed leaf peers
             idx2uuid = {g.get("index", i): g["uuid"]
                         for i, g in enumerate(gpus)}
-            for leaf, gpu in zip(leaves, gpus):
-                first_fill = not leaf.uuid
+
+            def fill(leaf, gpu, first_fill):
                 leaf.uuid = gpu["uuid"]
                 leaf.full_memory = int(gpu["memory"])
                 links = gpu.get("xgmi_links")
@@ -134,7 +134,38 @@ class CellTree:
                     leaf.available = 1.0
                     leaf.available_whole = 1
                 self.leaf_by_uuid[leaf.uuid] = leaf
+
+            # UUID-STABLE assignment: a re-register must keep each
+            # surviving GPU on the leaf that carries its reservations —
+            # positional assignment would re-bind every leaf when an
+            # earlier-indexed GPU drops out of the inventory.
+            by_uuid = {c.uuid: c for c in leaves if c.uuid}
+            new_gpus = []
+            for gpu in gpus:
+                leaf = by_uuid.get(gpu["uuid"])
+                if leaf is None:
+                    new_gpus.append(gpu)
+                else:
+                    fill(leaf, gpu, first_fill=False)
+            empty = [c for c in leaves if not c.uuid]
+            for leaf, gpu in zip(empty, new_gpus):
+                fill(leaf, gpu, first_fill=True)
             self.set_subtree_health(node_cell, healthy)
+            # per-GPU failure: a leaf whose GPU no longer appears in
+            # the inventory (device dropped off amdsmi) must not stay
+            # schedulable on a stale UUID (the reference only had
+            # node-level health, node.go:95-254)
+            present = {g["uuid"] for g in gpus}
+            missing = [c for c in leaves if c.uuid and
+                       c.uuid not in present]
+            for leaf in missing:
+                leaf.healthy = False
+            if missing:
+                p = node_cell
+                while p is not None:
+                    if p.children:
+                        p.healthy = any(ch.healthy for ch in p.children)
+                    p = p.parent
             self._recompute_up(node_cell)
 
     def set_subtree_health(self, cell: Cell, healthy: bool):

@@ -356,3 +356,30 @@ cells:
     fc.schedule_pending()
     nodes2 = {fc.pods[f"ns/s{i}"].node for i in (0, 1)}
     assert len(nodes2) == 1 and nodes2 != nodes
+
+
+def test_single_gpu_failure_uuid_stable():
+    """One GPU dropping out of the node inventory: its leaf goes
+    unhealthy, SURVIVING GPUs keep their leaves (and reservations) —
+    positional re-binding would scramble them."""
+    from kubeshare_amd.scheduler.plugin import KubeShareScheduler
+    sch = KubeShareScheduler(TopologyConfig.single_node("n", gpus=3))
+    inv = [{"uuid": f"G{i}", "memory": 1000, "index": i}
+           for i in range(3)]
+    sch.register_node("n", {C.MI355X_MODEL: inv})
+    # reserve half of G1
+    leaf1 = sch.tree.leaf_by_uuid["G1"]
+    sch.tree.reserve(leaf1, 0.5, 500)
+    # G0 dies; inventory now reports only G1, G2
+    sch.register_node("n", {C.MI355X_MODEL: inv[1:]})
+    assert sch.tree.leaf_by_uuid["G1"] is leaf1      # same leaf object
+    assert leaf1.available == pytest.approx(0.5)      # reservation kept
+    leaf0 = sch.tree.leaf_by_uuid["G0"]
+    assert not leaf0.healthy                          # failed GPU out
+    assert sch.tree.leaf_by_uuid["G2"].healthy
+    healthy = sch.tree.leaves_on_node("n")
+    assert {c.uuid for c in healthy} == {"G1", "G2"}
+    # G0 comes back: healthy again, reservations still intact on G1
+    sch.register_node("n", {C.MI355X_MODEL: inv})
+    assert sch.tree.leaf_by_uuid["G0"].healthy
+    assert leaf1.available == pytest.approx(0.5)
