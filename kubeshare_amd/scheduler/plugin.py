@@ -301,6 +301,26 @@ class KubeShareScheduler:
         picked = []
         remaining = spec.request
         memory = spec.memory
+        if need_whole and remaining > 1.0:
+            # greedy link-aware subset: after the first pick, prefer
+            # leaves directly xGMI-connected to what's already picked
+            # (a degraded link demotes that pair, same metric as gang
+            # locality) — RCCL rings over the chosen set stay on direct
+            # links where the topology allows it
+            pool = [c for _, c in scored]
+            base = {c.id: s for s, c in scored}
+            while pool and remaining > 0:
+                if not picked:
+                    c = pool.pop(0)
+                else:
+                    def key(c):
+                        loc = sum(self._distance(c, p.id) for p in picked)
+                        return base[c.id] - loc / len(picked) * 100.0
+                    c = max(pool, key=key)
+                    pool.remove(c)
+                picked.append(c)
+                remaining -= 1.0
+            return picked if remaining <= 0 else []
         for _, c in scored:
             if need_whole:
                 picked.append(c)

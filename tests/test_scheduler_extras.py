@@ -383,3 +383,24 @@ def test_single_gpu_failure_uuid_stable():
     sch.register_node("n", {C.MI355X_MODEL: inv})
     assert sch.tree.leaf_by_uuid["G0"].healthy
     assert leaf1.available == pytest.approx(0.5)
+
+
+def test_multi_gpu_pod_prefers_connected_subset():
+    """A single 2-GPU pod on a degraded topology picks a directly
+    linked pair, not the lowest indices."""
+    from kubeshare_amd.scheduler.harness import FakeCluster
+    from kubeshare_amd.scheduler.inventory import FakeInventory
+    fc = FakeCluster(nodes={"n": {"gpus": 4}})
+    inv = FakeInventory()
+    # GPU0 is islanded from 1 and 2; only 0-3 is a direct link
+    inv.add_node("n", gpus=4, down_links=[(0, 1), (0, 2)])
+    fc.inventory = inv
+    fc.scheduler.register_node("n", inv.by_model("n"))
+    pod = fc.add_pod("ns", "pair", {C.POD_GPU_REQUEST: "2.0",
+                                    C.POD_GPU_LIMIT: "2.0"})
+    fc.schedule_pending()
+    assert pod.phase == "Bound"
+    uuids = set(pod.annotations[C.POD_GPU_UUID].split(","))
+    # first pick is GPU-0 (tie -> list order); its partner must be the
+    # only direct neighbor, GPU-3
+    assert uuids == {"GPU-n-0", "GPU-n-3"}, uuids
