@@ -261,6 +261,21 @@ class FakeAPIServer:
                 _, _, parts = self._route()
                 n = int(self.headers.get("Content-Length", 0))
                 body = json.loads(self.rfile.read(n) or b"{}")
+                # /api/v1/namespaces/<ns>/pods/<name> merge patch
+                if len(parts) == 6 and parts[2] == "namespaces" and \
+                        parts[4] == "pods":
+                    with state.lock:
+                        pod = state.pods.get((parts[3], parts[5]))
+                        if pod is None:
+                            return self._send({"kind": "Status",
+                                               "code": 404}, 404)
+                        meta = body.get("metadata", {})
+                        pod.setdefault("metadata", {}).setdefault(
+                            "annotations", {}).update(
+                            meta.get("annotations") or {})
+                        pod["metadata"].setdefault("labels", {}).update(
+                            meta.get("labels") or {})
+                        return self._send(pod)
                 # /api/v1/nodes/<name> merge patch (inventory annotation)
                 if parts[:3] == ["api", "v1", "nodes"] and len(parts) == 4:
                     with state.lock:

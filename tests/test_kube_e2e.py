@@ -330,3 +330,13 @@ def test_apply_failure_rolls_back_reservation(cluster):
     for leaf in driver.sched.tree.leaves_on_node("node-a"):
         assert leaf.available == 1.0, leaf
     assert driver.sched.ports["node-a"].available()
+
+
+def test_pod_merge_patch_route(cluster):
+    srv, api, driver = cluster
+    srv.submit_pod("default", "patchme", {})
+    api.patch_namespaced_pod("patchme", "default", {
+        "metadata": {"annotations": {"x": "1"}, "labels": {"y": "2"}}})
+    pod = api.read_namespaced_pod("patchme", "default")
+    assert pod.metadata.annotations["x"] == "1"
+    assert pod.metadata.labels["y"] == "2"
