@@ -111,8 +111,20 @@ class KubeShareScheduler:
                 return None, None  # regular pod: not ours
             self.pod_status[key] = spec
         if spec.pod_group and spec.min_available > 0:
-            self.groups.get_or_create(namespace, spec.pod_group,
-                                      spec.priority, spec.min_available)
+            info = self.groups.get_or_create(namespace, spec.pod_group,
+                                             spec.priority,
+                                             spec.min_available)
+            # members of one gang must agree on minAvailable and
+            # priority (reference PreFilter scheduler.go:295-314
+            # rejects mismatches against the group's registered values)
+            if spec.min_available != info.min_available:
+                return None, (f"pod minAvailable {spec.min_available} "
+                              f"differs from group {spec.pod_group}'s "
+                              f"{info.min_available}")
+            if spec.priority != info.priority:
+                return None, (f"pod priority {spec.priority} differs "
+                              f"from group {spec.pod_group}'s "
+                              f"{info.priority}")
             total = (all_pods_in_group if all_pods_in_group is not None
                      else self._total_group_pods(namespace, spec.pod_group))
             if total < spec.min_available:

@@ -404,3 +404,29 @@ def test_multi_gpu_pod_prefers_connected_subset():
     # first pick is GPU-0 (tie -> list order); its partner must be the
     # only direct neighbor, GPU-3
     assert uuids == {"GPU-n-0", "GPU-n-3"}, uuids
+
+
+def test_gang_member_mismatch_rejected():
+    """Reference PreFilter parity (scheduler.go:295-314): a gang member
+    whose minAvailable or priority differs from the group's registered
+    values is rejected."""
+    from kubeshare_amd.scheduler.harness import FakeCluster
+    fc = FakeCluster(nodes={"n": {"gpus": 4}})
+    # group values are registered by the FIRST member through
+    # PreFilter (reference getOrCreatePodGroupInfo semantics); on the
+    # first cycle the harness queue orders unparsed pods by key, so
+    # name the well-formed members first alphabetically
+    base = {C.POD_GPU_REQUEST: "0.5", C.POD_GPU_LIMIT: "1.0",
+            C.POD_PRIORITY: "100", C.POD_GROUP_NAME: "g",
+            C.POD_GROUP_HEADCOUNT: "2", C.POD_GROUP_THRESHOLD: "1.0"}
+    fc.add_pod("ns", "a0", base)
+    fc.add_pod("ns", "a1", dict(base))
+    p_bad = fc.add_pod("ns", "zz-prio",
+                       dict(base, **{C.POD_PRIORITY: "50"}))
+    p_bad2 = fc.add_pod("ns", "zz-min",
+                        dict(base, **{C.POD_GROUP_HEADCOUNT: "3"}))
+    fc.schedule_pending()
+    assert p_bad.phase == "Unschedulable"
+    assert p_bad2.phase == "Unschedulable"
+    assert fc.pods["ns/a0"].phase == "Bound"
+    assert fc.pods["ns/a1"].phase == "Bound"
