@@ -116,15 +116,57 @@ class KubeDriver:
     # ---------------------------------------------------------- the loop
     def run_once(self):
         self.expire_permits()
-        pods = [p for p in self.v1.list_pod_for_all_namespaces(
-                    field_selector="status.phase=Pending").items
+        all_pods = self.v1.list_pod_for_all_namespaces().items
+        self.reconcile_deleted(all_pods)
+        pods = [p for p in all_pods
                 if p.spec.scheduler_name == C.SCHEDULER_NAME
-                and not p.spec.node_name]
+                and not p.spec.node_name
+                and (not p.status or p.status.phase in (None, "Pending"))]
         for p in sorted(pods, key=self._queue_key):
             if f"{p.metadata.namespace}/{p.metadata.name}" in \
                     self.waiting_pods:
                 continue  # parked at Permit; released or expired, not rerun
             self.schedule_pod(p)
+
+    def reconcile_deleted(self, all_pods):
+        """Reclaim reservations of pods that left the cluster or
+        completed (the reference's informer DeleteFunc / completed-pod
+        path, pod.go:91-136 — without this a long-running driver leaks
+        cell availability and manager ports until restart)."""
+        live = {}
+        for p in all_pods:
+            phase = p.status.phase if p.status else ""
+            if phase in ("Succeeded", "Failed"):
+                continue
+            key = f"{p.metadata.namespace}/{p.metadata.name}"
+            ann = p.metadata.annotations or {}
+            live[key] = (p.metadata.uid,
+                         ann.get("kubeshare.amd/original-uid", ""),
+                         p.spec.node_name or "")
+        for key, spec in list(self.sched.pod_status.items()):
+            if key in self.waiting_pods:
+                continue  # parked here; expiry handles it
+            cur = live.get(key)
+            if cur is not None:
+                uid, orig_uid, node = cur
+                # same key, SAME instance: the shadow-recreated pod has
+                # a new uid but carries the original-uid annotation
+                if uid == spec.uid or orig_uid == spec.uid or \
+                        not spec.uuids:
+                    continue
+                # same name but a NEW unbound instance (controller
+                # recreated it): the old instance's reservations must
+                # go before the new one is parsed over the spec
+                if node:
+                    continue  # bound namesake: resync owns this case
+            ns, name = key.split("/", 1)
+            self.sched.delete_pod(ns, name)
+            self.event_key(key, "Reclaimed",
+                           "pod gone/completed; resources reclaimed")
+
+    def event_key(self, key, reason, message):
+        print(f"[kubeshare-scheduler] {key}: {reason}: {message}",
+              flush=True)
 
     def _queue_key(self, pod):
         """Reference QueueSort (scheduler.go:247-267): group priority

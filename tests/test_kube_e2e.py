@@ -340,3 +340,57 @@ def test_pod_merge_patch_route(cluster):
     pod = api.read_namespaced_pod("patchme", "default")
     assert pod.metadata.annotations["x"] == "1"
     assert pod.metadata.labels["y"] == "2"
+
+
+def test_deleted_pod_reservations_reclaimed(cluster):
+    """A long-running driver must reclaim when bound pods are deleted
+    or complete (reference informer DeleteFunc, pod.go:91-136) — no
+    restart required."""
+    srv, api, driver = cluster
+    srv.submit_pod("default", "d1",
+                   {C.POD_GPU_REQUEST: "0.5", C.POD_GPU_LIMIT: "1.0"})
+    driver.run_once()
+    ann = dict(api.read_namespaced_pod("d1", "default")
+               .metadata.annotations.items())
+    leaf = driver.sched.tree.leaf_by_uuid[ann[C.POD_GPU_UUID]]
+    port = int(ann[C.POD_MANAGER_PORT])
+    assert leaf.available == pytest.approx(0.5)
+    # user deletes the running pod
+    api.delete_namespaced_pod("d1", "default")
+    driver.run_once()
+    assert leaf.available == pytest.approx(1.0)
+    assert driver.sched.ports["node-a"].is_free(port)
+    assert "default/d1" not in driver.sched.pod_status
+
+    # completion (phase Succeeded) reclaims too
+    srv.submit_pod("default", "d2",
+                   {C.POD_GPU_REQUEST: "0.5", C.POD_GPU_LIMIT: "1.0"})
+    driver.run_once()
+    with srv.lock:
+        srv.pods[("default", "d2")]["status"]["phase"] = "Succeeded"
+    driver.run_once()
+    for lf in driver.sched.tree.leaves_on_node("node-a"):
+        assert lf.available == pytest.approx(1.0)
+
+
+def test_namesake_recreation_reclaims_old_instance(cluster):
+    """A controller deleting a bound pod and recreating the NAME with a
+    new uid: the old instance's reservation is reclaimed before the new
+    Pending instance schedules (no double-charge, no leak)."""
+    srv, api, driver = cluster
+    srv.submit_pod("default", "twin",
+                   {C.POD_GPU_REQUEST: "0.5", C.POD_GPU_LIMIT: "1.0"})
+    driver.run_once()
+    # controller replaces it: same name, fresh uid, Pending, unbound
+    with srv.lock:
+        srv.pods.pop(("default", "twin"))
+    srv.submit_pod("default", "twin",
+                   {C.POD_GPU_REQUEST: "0.5", C.POD_GPU_LIMIT: "1.0"})
+    driver.run_once()
+    # exactly ONE 0.5 reservation exists afterwards
+    total_reserved = sum(1.0 - lf.available
+                         for lf in driver.sched.tree.leaves_on_node(
+                             "node-a"))
+    assert total_reserved == pytest.approx(0.5)
+    pod = api.read_namespaced_pod("twin", "default")
+    assert pod.spec.node_name == "node-a"
