@@ -63,9 +63,11 @@ class KubeDriver:
 
     # ------------------------------------------------------------ nodes
     def sync_nodes(self):
+        seen = set()
         for node in self.v1.list_node(
                 label_selector="SharedGPU=true").items:
             name = node.metadata.name
+            seen.add(name)
             if self.inventory is not None:
                 by_model = self.inventory.by_model(name)
             else:
@@ -96,6 +98,12 @@ class KubeDriver:
                 self.sched.register_node(name, by_model, healthy=ready)
             else:
                 self.sched.set_node_health(name, ready)
+        # a topology node that lost its SharedGPU label or left the
+        # cluster must stop receiving pods (reference updateNode label
+        # handling, node.go:54-68)
+        for n in self.sched.tree.all_nodes():
+            if n not in seen:
+                self.sched.set_node_health(n, False)
 
     # ------------------------------------------------------- resync
     def resync_bound(self):

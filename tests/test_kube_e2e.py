@@ -394,3 +394,25 @@ def test_namesake_recreation_reclaims_old_instance(cluster):
     assert total_reserved == pytest.approx(0.5)
     pod = api.read_namespaced_pod("twin", "default")
     assert pod.spec.node_name == "node-a"
+
+
+def test_unlabeled_node_stops_receiving_pods(cluster):
+    """Removing SharedGPU=true (or the node leaving the cluster) makes
+    its cells unhealthy on the next node sync (reference updateNode,
+    node.go:54-68)."""
+    srv, api, driver = cluster
+    with srv.lock:
+        srv.nodes["node-a"]["metadata"]["labels"].pop("SharedGPU")
+    driver.sync_nodes()
+    srv.submit_pod("default", "nohome",
+                   {C.POD_GPU_REQUEST: "0.5", C.POD_GPU_LIMIT: "1.0"})
+    driver.run_once()
+    pod = api.read_namespaced_pod("nohome", "default")
+    assert not pod.spec.node_name          # nowhere to go
+    # relabel -> schedulable again
+    with srv.lock:
+        srv.nodes["node-a"]["metadata"]["labels"]["SharedGPU"] = "true"
+    driver.sync_nodes()
+    driver.run_once()
+    pod = api.read_namespaced_pod("nohome", "default")
+    assert pod.spec.node_name == "node-a"
