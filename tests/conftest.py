@@ -13,6 +13,9 @@ sys.path.insert(0, REPO)
 def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: test requires a real MI355X (run via gpurun)")
+    config.addinivalue_line(
+        "markers", "slow: multi-minute load test (deselect with -m "
+                   "'not slow' for quick iterations)")
 
 
 def _built(name: str) -> str:
