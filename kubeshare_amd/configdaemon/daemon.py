@@ -89,7 +89,8 @@ class PrometheusPodSource:
                     memory=int(s.get("memory", 0)),
                     port=int(s.get("port", 0)),
                     group_name=s.get("group_name", ""),
-                    cell_id=s.get("cell_id", "")))
+                    cell_id=s.get("cell_id", ""),
+                    lease_ms=int(s.get("lease_ms", 0) or 0)))
             except (ValueError, TypeError):
                 continue
         return out
