@@ -43,7 +43,11 @@ def check_hook_active(required: bool):
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--model", default="resnet50")
+    ap.add_argument("--model", default="resnet50",
+                    choices=["resnet18", "resnet50", "vgg16"],
+                    help="ImageNet-shaped families only (the mnist/"
+                         "lstm families have their own input shapes — "
+                         "see models.small.synthetic_batch)")
     ap.add_argument("--batch", type=int, default=64)
     ap.add_argument("--image-size", type=int, default=224)
     ap.add_argument("--steps", type=int, default=20)
