@@ -6,4 +6,4 @@ HIP LD_PRELOAD isolation (native/), xGMI-topology-aware scheduling
 daemons (configdaemon/, noded/), and hand-written CDNA4 kernels (ops/).
 See README.md and PARITY.md.
 """
-__version__ = "0.1.0"
+__version__ = "0.2.0"
