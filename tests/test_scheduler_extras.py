@@ -430,3 +430,16 @@ def test_gang_member_mismatch_rejected():
     assert p_bad2.phase == "Unschedulable"
     assert fc.pods["ns/a0"].phase == "Bound"
     assert fc.pods["ns/a1"].phase == "Bound"
+
+
+def test_synthetic_batch_shapes_match_models():
+    import torch
+    from kubeshare_amd.models import build_model
+    from kubeshare_amd.models.small import synthetic_batch
+    for name in ("mnist", "lstm"):
+        m = build_model(name)
+        x, y = synthetic_batch(m, 4)
+        out = m(x)
+        assert out.shape == (4, 10)
+        assert y.shape == (4,) and int(y.max()) < 10
+        assert torch.isfinite(out).all()
