@@ -100,7 +100,7 @@ int main(int argc, char** argv) {
     }
     std::vector<PodQuota> quotas;
     if (load_gpu_config(config_path.c_str(), quotas)) {
-      sched.set_config(quotas);
+      sched.set_config(quotas, now_ms());
       logf(g_log, "gpu-schd", "loaded %zu pod quotas from %s", quotas.size(),
            config_path.c_str());
     }
@@ -195,7 +195,7 @@ int main(int argc, char** argv) {
       if (ours && !config_path.empty()) {
         std::vector<PodQuota> quotas;
         if (load_gpu_config(config_path.c_str(), quotas)) {
-          sched.set_config(quotas);
+          sched.set_config(quotas, now_ms());
           logf(g_log, "gpu-schd", "config reloaded: %zu pods", quotas.size());
         }
       }
