@@ -467,6 +467,30 @@ int main() {
     CHECK(out[3].memory == 7 && out[3].request == 0.4);
   }
 
+  // ---- 21. departed-pod account pruning (live-clock set_config)
+  {
+    TokenScheduler s(50, 10, 1000);
+    s.set_config({quota("old", 1.0, 0.5)}, 0.0);
+    Grant g;
+    double retry;
+    s.request("old", 1, 0, 0.0);
+    CHECK(s.schedule(0.0, &g, &retry));
+    s.release("old", 50, 50.0);
+    // pod leaves the config while its usage is still in the window:
+    // account is retained (anti-gaming: usage must keep counting)
+    s.set_config({quota("new", 1.0, 0.5)}, 100.0);
+    CHECK(s.stats_json(100.0).find("\"old\"") != std::string::npos);
+    // after the window ages the charge out, the account is pruned
+    s.set_config({quota("new", 1.0, 0.5)}, 2000.0);
+    CHECK(s.stats_json(2000.0).find("\"old\"") == std::string::npos);
+    // a pod still HOLDING is never pruned even out-of-config
+    s.request("new", 2, 0, 2000.0);
+    CHECK(s.schedule(2000.0, &g, &retry));
+    s.set_config({}, 9999999.0);
+    CHECK(s.n_holders() == 1);
+    CHECK(s.stats_json(9999999.0).find("\"new\"") != std::string::npos);
+  }
+
   printf("sched_test OK (%d checks)\n", g_checks);
   return 0;
 }
