@@ -421,6 +421,7 @@ def test_pod_mgr_uds_default_transport(tmp_path, native_bins, fake_gpus):
 
 
 @pytest.mark.slow
+@pytest.mark.timeout(1500)  # ~200 s here; headroom for slower CI boxes
 def test_simulator_reference_trace_full():
     """The reference's actual 989-job arrival trace
     (test/simulator/trace.txt) replayed end to end against the
