@@ -436,7 +436,7 @@ def test_simulator_reference_trace_full():
     r = subprocess.run(
         [_sys.executable, "tools/simulator.py", "--trace", trace,
          "--nodes", "4"],
-        cwd=repo, capture_output=True, text=True, timeout=600)
+        cwd=repo, capture_output=True, text=True, timeout=1400)
     assert r.returncode == 0, r.stderr[-1500:]
     out = json.loads(r.stdout.strip().splitlines()[-1])
     assert out["jobs"] == 989
