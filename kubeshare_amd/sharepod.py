@@ -20,6 +20,7 @@ _FIELD_TO_LABEL = {
     "gpuMem": C.POD_GPU_MEMORY,
     "gpuModel": C.POD_GPU_MODEL,
     "priority": C.POD_PRIORITY,
+    "leaseMs": C.POD_LEASE_MS,
     "groupName": C.POD_GROUP_NAME,
     "groupHeadcount": C.POD_GROUP_HEADCOUNT,
     "groupThreshold": C.POD_GROUP_THRESHOLD,
