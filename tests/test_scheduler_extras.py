@@ -443,3 +443,13 @@ def test_synthetic_batch_shapes_match_models():
         assert out.shape == (4, 10)
         assert y.shape == (4,) and int(y.max()) < 10
         assert torch.isfinite(out).all()
+
+
+def test_port_pool_is_free_bounds():
+    pool = RRPortPool(base=50050, size=4)
+    p = pool.allocate()
+    assert not pool.is_free(p)
+    pool.release(p)
+    assert pool.is_free(p)
+    assert not pool.is_free(50049)   # out of range is never "free"
+    assert not pool.is_free(50054)
