@@ -453,3 +453,47 @@ def test_port_pool_is_free_bounds():
     assert pool.is_free(p)
     assert not pool.is_free(50049)   # out of range is never "free"
     assert not pool.is_free(50054)
+
+
+def test_mixed_models_on_one_host():
+    """One host carrying two GPU models (two node-level cells with the
+    same node name, reference multi-chain config): model pinning picks
+    the right leaves; unpinned pods prefer the higher-priority model."""
+    from kubeshare_amd.scheduler.harness import FakeCluster
+    from kubeshare_amd.scheduler.plugin import KubeShareScheduler
+    from kubeshare_amd.scheduler.topology import (CellSpec, CellTypeSpec,
+                                                  TopologyConfig)
+    topo = TopologyConfig(
+        cell_types={
+            "FAST-NODE": CellTypeSpec("AMD Instinct MI355X", 2, 100, True),
+            "SLOW-NODE": CellTypeSpec("AMD Instinct MI300X", 2, 50, True),
+        },
+        cells=[CellSpec(cell_type="FAST-NODE", cell_id="host1"),
+               CellSpec(cell_type="SLOW-NODE", cell_id="host1")])
+    fc = FakeCluster.__new__(FakeCluster)
+    fc.scheduler = KubeShareScheduler(topo)
+    fc.pods = {}
+    fc.clock = 0.0
+    fc.events = []
+    import itertools
+    fc._uid = itertools.count(1)
+    inv = {
+        "AMD Instinct MI355X": [
+            {"uuid": f"F{i}", "memory": 10_000, "index": i}
+            for i in range(2)],
+        "AMD Instinct MI300X": [
+            {"uuid": f"S{i}", "memory": 5_000, "index": 2 + i}
+            for i in range(2)],
+    }
+    fc.scheduler.register_node("host1", inv)
+    pinned = fc.add_pod("ns", "slowpod", {
+        C.POD_GPU_REQUEST: "0.5", C.POD_GPU_LIMIT: "1.0",
+        C.POD_GPU_MODEL: "AMD Instinct MI300X"})
+    fast = fc.add_pod("ns", "anypod", {
+        C.POD_GPU_REQUEST: "0.5", C.POD_GPU_LIMIT: "1.0"})
+    fc.schedule_pending()
+    assert pinned.phase == "Bound"
+    assert pinned.annotations[C.POD_GPU_UUID].startswith("S")
+    assert fast.phase == "Bound"
+    # unpinned pod prefers the higher-priority (faster) model
+    assert fast.annotations[C.POD_GPU_UUID].startswith("F")
