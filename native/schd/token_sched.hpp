@@ -357,12 +357,20 @@ class TokenScheduler {
       kv.second.prune(now, window_);
       total_busy += kv.second.usage_cache;
     }
+    std::string holders;
+    for (auto& kv : holders_) {
+      if (!holders.empty()) holders += ",";
+      holders += "\"" + kv.first + "\":" +
+                 fmt(now - kv.second.grant_ms);
+    }
     std::string s = "{\"window_ms\":" + fmt(window_) +
                     ",\"busy_ms\":" + fmt(total_busy) +
                     ",\"sampler\":" + (sampler_active_ ? "true" : "false") +
                     ",\"other_busy_ms\":" + fmt(other_busy_ms_) +
                     ",\"revokes\":" + std::to_string(revokes_) +
                     ",\"last_revoked\":\"" + last_revoked_ + "\"" +
+                    ",\"holders\":{" + holders + "}" +
+                    ",\"waiters\":" + std::to_string(waiters_.size()) +
                     ",\"pods\":{";
     bool first = true;
     for (auto& kv : pods_) {
