@@ -43,6 +43,13 @@ class GpuSchdCollector:
             "gpu_schd_unattributed_busy_ms",
             "sampled GPU-busy with no token holder (exempt RCCL "
             "kernels, ungated processes)", labels=["node", "uuid"])
+        waiters = GaugeMetricFamily(
+            "gpu_schd_waiters", "token requests currently queued",
+            labels=["node", "uuid"])
+        revokes = CounterMetricFamily(
+            "gpu_schd_revokes",
+            "liveness revocations (holder died or hung)",
+            labels=["node", "uuid"])
         for uuid, (host, port) in self.endpoints.items():
             try:
                 st = query_stats(host, port, timeout=2.0)
@@ -52,6 +59,10 @@ class GpuSchdCollector:
                                1.0 if st.get("sampler") else 0.0)
             other.add_metric([self.node_name, uuid],
                              st.get("other_busy_ms", 0.0))
+            waiters.add_metric([self.node_name, uuid],
+                               st.get("waiters", 0))
+            revokes.add_metric([self.node_name, uuid],
+                               st.get("revokes", 0))
             for pod, v in st.get("pods", {}).items():
                 lab = [self.node_name, uuid, pod]
                 usage.add_metric(lab, v["usage_ms"])
@@ -64,6 +75,8 @@ class GpuSchdCollector:
         yield grants
         yield sampler
         yield other
+        yield waiters
+        yield revokes
 
 
 def serve(endpoints: dict, node_name: str, port: int):
