@@ -95,6 +95,9 @@ class SharePodController:
                     or "Pending"
                 node = pod.spec.node_name or ""
             status = {"phase": phase, "node": node, "podName": name}
+            cur = obj.get("status") or {}
+            if all(cur.get(k) == v for k, v in status.items()):
+                continue  # steady state: no no-op write every cycle
             self.crd.patch_namespaced_custom_object_status(
                 self.GROUP, self.VERSION, ns, self.PLURAL, name,
                 {"status": status})

@@ -164,6 +164,14 @@ def test_sharepod_crd_end_to_end(cluster):
     obj = crd.get_namespaced_custom_object(G, V, "default", P, "sp1")
     assert obj["status"]["phase"] == "Running"
     assert obj["status"]["node"] == "node-a"
+    # steady state: further reconciles are read-only (no no-op status
+    # writes from the 2s control loop)
+    writes = []
+    orig = crd.patch_namespaced_custom_object_status
+    crd.patch_namespaced_custom_object_status = \
+        lambda *a, **kw: writes.append(a) or orig(*a, **kw)
+    ctl.reconcile_once()
+    assert writes == []
 
 
 def test_noded_publishes_inventory_annotation(cluster, tmp_path):
