@@ -53,7 +53,9 @@ class GpuSchdCollector:
         for uuid, (host, port) in self.endpoints.items():
             try:
                 st = query_stats(host, port, timeout=2.0)
-            except OSError:
+            except (OSError, ValueError):
+                # daemon down, or truncated/garbled STATS mid-restart:
+                # skip this GPU, never fail the whole node scrape
                 continue
             sampler.add_metric([self.node_name, uuid],
                                1.0 if st.get("sampler") else 0.0)

@@ -282,6 +282,56 @@ def test_noded_metrics_exporter(tmp_path, native_bins, fake_gpus):
         nd.stop()
 
 
+def test_noded_metrics_skips_garbled_endpoint():
+    """A gpu-schd replying garbage (mid-restart truncation) must not
+    fail the whole node scrape — that endpoint is skipped."""
+    import threading
+    from kubeshare_amd.noded.metrics import GpuSchdCollector
+
+    srv = socket.socket()
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(4)
+    port = srv.getsockname()[1]
+    stop = threading.Event()
+
+    def junk_server():
+        srv.settimeout(0.2)
+        while not stop.is_set():
+            try:
+                c, _ = srv.accept()
+            except OSError:
+                continue
+            try:
+                c.recv(256)
+                c.sendall(b"not json {{{\n")
+            finally:
+                c.close()
+
+    t = threading.Thread(target=junk_server, daemon=True)
+    t.start()
+    try:
+        reg = CollectorRegistry()
+        reg.register(GpuSchdCollector(
+            {"GPU-bad": ("127.0.0.1", port),
+             "GPU-down": ("127.0.0.1", free_port_unused())}, "node-a"))
+        text = generate_latest(reg).decode()
+        # scrape survives; families render with no samples for bad GPUs
+        assert "gpu_pod_window_usage_ms" in text
+        assert "GPU-bad" not in text and "GPU-down" not in text
+    finally:
+        stop.set()
+        t.join(timeout=2)
+        srv.close()
+
+
+def free_port_unused():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
 def test_pod_group_gc():
     import time as _t
     from kubeshare_amd.scheduler.pod_group import PodGroupRegistry
