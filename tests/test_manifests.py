@@ -49,3 +49,19 @@ def test_crd_matches_sharepod_controller():
     assert crd["spec"]["names"]["plural"] == sp.SharePodController.PLURAL
     served = [v["name"] for v in crd["spec"]["versions"] if v["served"]]
     assert sp.SharePodController.VERSION in served
+
+
+def test_console_scripts_resolve():
+    """Every [project.scripts] entry point must import and be callable
+    (packaging bitrot guard)."""
+    import importlib
+
+    import tomli
+
+    with open(os.path.join(REPO, "pyproject.toml"), "rb") as f:
+        cfg = tomli.load(f)
+    scripts = cfg["project"]["scripts"]
+    assert len(scripts) >= 3
+    for target in scripts.values():
+        mod, fn = target.split(":")
+        assert callable(getattr(importlib.import_module(mod), fn)), target
