@@ -43,24 +43,27 @@ def demand_from_pod(pod) -> PodDemand | None:
     if C.POD_GPU_UUID not in ann:
         return None
     try:
-        limit = float(labels.get(C.POD_GPU_LIMIT, "0") or 0)
-        request = float(labels.get(C.POD_GPU_REQUEST, "0") or 0)
+        # scheduler-written annotations are trusted, but a hand-crafted
+        # pod can carry junk in any numeric field — one bad pod must
+        # not fail the whole cluster scrape
+        return PodDemand(
+            namespace=pod.namespace, name=pod.name,
+            pod_id=getattr(pod, "uid", ""),
+            node=getattr(pod, "node", "") or getattr(pod, "node_name",
+                                                     ""),
+            uuid=ann.get(C.POD_GPU_UUID, ""),
+            limit=float(labels.get(C.POD_GPU_LIMIT, "0") or 0),
+            request=float(labels.get(C.POD_GPU_REQUEST, "0") or 0),
+            memory=int(ann.get(C.POD_GPU_MEMORY, "0") or 0),
+            port=int(ann.get(C.POD_MANAGER_PORT, "0") or
+                     env.get(C.ENV_POD_MANAGER_PORT, "0") or 0),
+            group_name=labels.get(C.POD_GROUP_NAME, ""),
+            min_available=int(labels.get(C.POD_MIN_AVAILABLE, "0") or 0),
+            cell_id=ann.get(C.POD_CELL_ID, ""),
+            lease_ms=int(labels.get(C.POD_LEASE_MS, "0") or 0),
+        )
     except ValueError:
         return None
-    return PodDemand(
-        namespace=pod.namespace, name=pod.name,
-        pod_id=getattr(pod, "uid", ""),
-        node=getattr(pod, "node", "") or getattr(pod, "node_name", ""),
-        uuid=ann.get(C.POD_GPU_UUID, ""),
-        limit=limit, request=request,
-        memory=int(ann.get(C.POD_GPU_MEMORY, "0") or 0),
-        port=int(ann.get(C.POD_MANAGER_PORT, "0") or
-                 env.get(C.ENV_POD_MANAGER_PORT, "0") or 0),
-        group_name=labels.get(C.POD_GROUP_NAME, ""),
-        min_available=int(labels.get(C.POD_MIN_AVAILABLE, "0") or 0),
-        cell_id=ann.get(C.POD_CELL_ID, ""),
-        lease_ms=int(labels.get(C.POD_LEASE_MS, "0") or 0),
-    )
 
 
 class GPURequirementCollector:

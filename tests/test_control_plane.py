@@ -282,6 +282,23 @@ def test_noded_metrics_exporter(tmp_path, native_bins, fake_gpus):
         nd.stop()
 
 
+def test_demand_from_pod_rejects_junk_numerics():
+    """A hand-crafted pod with non-numeric gpu_mem/port/lease_ms must
+    yield None, not crash the aggregator scrape."""
+    class P:
+        namespace, name, uid, node = "ns", "bad", "u", "n"
+        labels = {C.POD_GPU_LIMIT: "1.0", C.POD_GPU_REQUEST: "0.5",
+                  C.POD_LEASE_MS: "fast"}
+        annotations = {C.POD_GPU_UUID: "GPU-x", C.POD_GPU_MEMORY: "lots"}
+        env = {}
+
+    assert demand_from_pod(P()) is None
+    P.labels[C.POD_LEASE_MS] = "25"
+    P.annotations[C.POD_GPU_MEMORY] = "1024"
+    d = demand_from_pod(P())
+    assert d is not None and d.lease_ms == 25 and d.memory == 1024
+
+
 def test_noded_metrics_skips_garbled_endpoint():
     """A gpu-schd replying garbage (mid-restart truncation) must not
     fail the whole node scrape — that endpoint is skipped."""
