@@ -178,3 +178,29 @@ def test_webhook_patch_idempotent_property(containers, port, volumes):
         assert C.ENV_INJECTED in names
         if port:
             assert names.count("POD_MANAGER_UDS") == 1
+
+
+# ---------------------------------------------------------------- aggregator
+_junk = st.text(string.printable, max_size=12)
+
+
+@given(labels=st.dictionaries(st.sampled_from([
+           C.POD_GPU_LIMIT, C.POD_GPU_REQUEST, C.POD_MIN_AVAILABLE,
+           C.POD_LEASE_MS, C.POD_GROUP_NAME]), _junk, max_size=5),
+       ann=st.dictionaries(st.sampled_from([
+           C.POD_GPU_UUID, C.POD_GPU_MEMORY, C.POD_MANAGER_PORT,
+           C.POD_CELL_ID]), _junk, max_size=4))
+@settings(max_examples=200, deadline=None)
+def test_demand_from_pod_never_raises(labels, ann):
+    """Scrape safety: arbitrary junk in labels/annotations yields a
+    PodDemand or None, never an exception (one bad pod must not 500
+    the aggregator endpoint)."""
+    from kubeshare_amd.aggregator import PodDemand, demand_from_pod
+
+    class P:
+        namespace, name, uid, node = "ns", "p", "u", "n"
+        env = {}
+
+    P.labels, P.annotations = labels, ann
+    d = demand_from_pod(P())
+    assert d is None or isinstance(d, PodDemand)
