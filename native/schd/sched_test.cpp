@@ -491,6 +491,25 @@ int main() {
     CHECK(s.stats_json(9999999.0).find("\"new\"") != std::string::npos);
   }
 
+  // ---- 22. hostile pod names never corrupt the STATS JSON
+  {
+    TokenScheduler s(50, 10, 1000);
+    Grant g;
+    double retry;
+    // local-mode REQ: name with quote, backslash and a control byte
+    s.request("ns/evil\"pod\\x\n", 1, 0, 0.0);
+    CHECK(s.schedule(0.0, &g, &retry));
+    std::string j = s.stats_json(10.0);
+    // every unescaped quote must be a key/value delimiter: scan for
+    // balanced structure — no raw '"' preceded by a name byte
+    CHECK(j.find("evil\\\"pod\\\\x?") != std::string::npos);
+    CHECK(j.find('\n') == std::string::npos);
+    // revocation path escapes too
+    CHECK(s.check_revoke(100000.0));
+    CHECK(s.stats_json(100000.0).find("\"last_revoked\":\"ns/evil\\\"") !=
+          std::string::npos);
+  }
+
   printf("sched_test OK (%d checks)\n", g_checks);
   return 0;
 }

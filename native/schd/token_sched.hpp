@@ -360,7 +360,7 @@ class TokenScheduler {
     std::string holders;
     for (auto& kv : holders_) {
       if (!holders.empty()) holders += ",";
-      holders += "\"" + kv.first + "\":" +
+      holders += "\"" + jesc(kv.first) + "\":" +
                  fmt(now - kv.second.grant_ms);
     }
     std::string s = "{\"window_ms\":" + fmt(window_) +
@@ -368,7 +368,7 @@ class TokenScheduler {
                     ",\"sampler\":" + (sampler_active_ ? "true" : "false") +
                     ",\"other_busy_ms\":" + fmt(other_busy_ms_) +
                     ",\"revokes\":" + std::to_string(revokes_) +
-                    ",\"last_revoked\":\"" + last_revoked_ + "\"" +
+                    ",\"last_revoked\":\"" + jesc(last_revoked_) + "\"" +
                     ",\"holders\":{" + holders + "}" +
                     ",\"waiters\":" + std::to_string(waiters_.size()) +
                     ",\"pods\":{";
@@ -379,7 +379,7 @@ class TokenScheduler {
       first = false;
       double share = total_busy > 0 ? a.usage_cache / total_busy : 0.0;
       double frac = a.usage_cache / window_;
-      s += "\"" + kv.first + "\":{\"request\":" + fmt(a.request) +
+      s += "\"" + jesc(kv.first) + "\":{\"request\":" + fmt(a.request) +
            ",\"limit\":" + fmt(a.limit) + ",\"usage_ms\":" +
            fmt(a.usage_cache) + ",\"window_frac\":" + fmt(frac) +
            ",\"busy_share\":" + fmt(share) + ",\"grants\":" +
@@ -396,6 +396,20 @@ class TokenScheduler {
     char b[32];
     snprintf(b, sizeof(b), "%.4f", v);
     return b;
+  }
+
+  // Pod names are k8s-validated in the cluster path, but local-mode
+  // REQs (hook loopback, no pod-mgr stamping) carry arbitrary bytes —
+  // they must not corrupt the STATS JSON reply.
+  static std::string jesc(const std::string& s) {
+    std::string out;
+    out.reserve(s.size());
+    for (char c : s) {
+      if (c == '"' || c == '\\') out += '\\';
+      if ((unsigned char)c < 0x20) { out += '?'; continue; }
+      out += c;
+    }
+    return out;
   }
 
   struct Holder {
