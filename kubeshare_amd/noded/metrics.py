@@ -66,11 +66,13 @@ class GpuSchdCollector:
             revokes.add_metric([self.node_name, uuid],
                                st.get("revokes", 0))
             for pod, v in st.get("pods", {}).items():
+                # .get(): a gpu-schd from an older build mid-rolling-
+                # upgrade may lack newer per-pod fields
                 lab = [self.node_name, uuid, pod]
-                usage.add_metric(lab, v["usage_ms"])
-                share.add_metric(lab, v["busy_share"])
-                total.add_metric(lab, v["total_used_ms"])
-                grants.add_metric(lab, v["grants"])
+                usage.add_metric(lab, v.get("usage_ms", 0.0))
+                share.add_metric(lab, v.get("busy_share", 0.0))
+                total.add_metric(lab, v.get("total_used_ms", 0.0))
+                grants.add_metric(lab, v.get("grants", 0))
         yield usage
         yield share
         yield total
