@@ -327,10 +327,11 @@ def test_noded_metrics_skips_garbled_endpoint():
     t = threading.Thread(target=junk_server, daemon=True)
     t.start()
     try:
+        from kubeshare_amd.isolation.local import free_port
         reg = CollectorRegistry()
         reg.register(GpuSchdCollector(
             {"GPU-bad": ("127.0.0.1", port),
-             "GPU-down": ("127.0.0.1", free_port_unused())}, "node-a"))
+             "GPU-down": ("127.0.0.1", free_port())}, "node-a"))
         text = generate_latest(reg).decode()
         # scrape survives; families render with no samples for bad GPUs
         assert "gpu_pod_window_usage_ms" in text
@@ -339,14 +340,6 @@ def test_noded_metrics_skips_garbled_endpoint():
         stop.set()
         t.join(timeout=2)
         srv.close()
-
-
-def free_port_unused():
-    s = socket.socket()
-    s.bind(("127.0.0.1", 0))
-    p = s.getsockname()[1]
-    s.close()
-    return p
 
 
 def test_pod_group_gc():
